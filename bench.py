@@ -1,0 +1,171 @@
+#!/usr/bin/env python3
+"""bench.py — flagship benchmark of the MI355X K3S enablement stack.
+
+Measures the stack's headline metric (BASELINE.json): **HIP STREAM triad
+GB/s** on 1..N MI355X GPUs — the GPU payload that our smoke pod runs where
+the reference stack ran `nvidia-smi` (/root/reference/nvidia-smi.yaml:13;
+the reference publishes no GB/s-class number, so vs_baseline is null and
+the measured value becomes the recorded baseline).
+
+One "step" = one STREAM triad sweep `a = b + s*c` over three fp32 buffers
+of --buffer-mib MiB each (bytes moved per step = 3 x buffer bytes, STREAM
+convention). Multi-GPU runs are launched by torch.distributed.run with one
+rank per GPU; each rank owns its own buffers (weak scaling) and the
+reported value is the whole-job aggregate GB/s computed with the MAX
+elapsed time over ranks.
+
+Contract (driver): rank 0 prints exactly one JSON line.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+
+def parse_args(argv=None):
+    p = argparse.ArgumentParser(description=__doc__)
+    p.add_argument("--gpus", type=int, default=None,
+                   help="number of GPUs (defaults to WORLD_SIZE or 1)")
+    p.add_argument("--steps", type=int, default=200)
+    p.add_argument("--warmup", type=int, default=50)
+    p.add_argument("--buffer-mib", type=int, default=1024,
+                   help="MiB per buffer (3 buffers per GPU)")
+    p.add_argument("--variant", choices=["auto", "plain", "nt"], default="auto",
+                   help="plain or non-temporal loads/stores (auto measures both"
+                        " during warmup and picks the faster globally)")
+    p.add_argument("--scalar", type=float, default=2.5)
+    return p.parse_args(argv)
+
+
+def build_result(*, value, n_gpus, steps, warmup, ms_per_step, buffer_mib,
+                 variant):
+    """Assemble the contract JSON (separated out for CPU-side tests)."""
+    return {
+        "metric": "hip_stream_triad_gbps",
+        "value": round(value, 1),
+        "unit": "GB/s",
+        "n_gpus": n_gpus,
+        "steps": steps,
+        "warmup": warmup,
+        "ms_per_step": round(ms_per_step, 4),
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,  # reference publishes no bandwidth number
+        "dtype": "fp32",
+        "data": "synthetic",
+        "config": {
+            "model": "hip_stream_triad",
+            "buffer_MiB": buffer_mib,
+            "n_buffers": 3,
+            "variant": variant,
+            "parallelism": f"dp{n_gpus}",
+        },
+    }
+
+
+def main(argv=None) -> int:
+    args = parse_args(argv)
+    import torch
+
+    world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    n_gpus = args.gpus or world_size
+
+    if not torch.cuda.is_available():
+        print("bench.py requires a GPU (run on the MI355X box)",
+              file=sys.stderr)
+        return 1
+
+    dist = None
+    if world_size > 1:
+        import torch.distributed as dist_mod
+        dist = dist_mod
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29571")
+        dist.init_process_group(backend="nccl", rank=rank,
+                                world_size=world_size)
+
+    torch.cuda.set_device(local_rank)
+    device = torch.device("cuda", local_rank)
+
+    from k3samd import ops
+    if not ops.native_available():
+        print("k3samd native extension missing — refusing to run a fallback",
+              file=sys.stderr)
+        return 2
+
+    n = args.buffer_mib * (1 << 20) // 4  # fp32 elements per buffer
+    b = torch.rand(n, device=device)
+    c = torch.rand(n, device=device)
+    a = torch.empty_like(b)
+    buffer_bytes = n * 4
+    step_bytes = 3 * buffer_bytes
+
+    def run_step(nt: bool):
+        ops.stream_triad(a, b, c, args.scalar, nontemporal=nt)
+
+    def barrier():
+        if dist is not None:
+            dist.barrier()
+
+    # ---- variant selection (globally consistent across ranks) ----
+    if args.variant == "auto":
+        timings = []
+        for nt in (False, True):
+            for _ in range(5):
+                run_step(nt)
+            torch.cuda.synchronize()
+            t0 = time.perf_counter()
+            for _ in range(10):
+                run_step(nt)
+            torch.cuda.synchronize()
+            timings.append(time.perf_counter() - t0)
+        t = torch.tensor(timings, dtype=torch.float64)
+        if dist is not None:
+            dist.all_reduce(t)  # sum over ranks -> same decision everywhere
+        use_nt = bool(t[1] < t[0])
+    else:
+        use_nt = args.variant == "nt"
+    variant = "nt" if use_nt else "plain"
+
+    # ---- warmup ----
+    for _ in range(args.warmup):
+        run_step(use_nt)
+    torch.cuda.synchronize()
+
+    # ---- timed region ----
+    barrier()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        run_step(use_nt)
+    torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+    barrier()
+
+    t = torch.tensor([elapsed], dtype=torch.float64)
+    if dist is not None:
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    t_max = float(t[0])
+
+    value = n_gpus * step_bytes * args.steps / t_max / 1e9
+    ms_per_step = t_max / args.steps * 1e3
+
+    if rank == 0:
+        res = build_result(value=value, n_gpus=n_gpus, steps=args.steps,
+                           warmup=args.warmup, ms_per_step=ms_per_step,
+                           buffer_mib=args.buffer_mib, variant=variant)
+        print(json.dumps(res))
+
+    if dist is not None:
+        dist.destroy_process_group()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -1,0 +1,60 @@
+"""CDNA4 HIP kernel bindings.
+
+The native extension (``k3samd/_C*.so``) is compiled in-tree for gfx950 by
+``k3samd.build``. On a GPU machine a missing extension is a hard error —
+there is deliberately no silent eager/PyTorch fallback for the compute path
+(the kernels ARE the product, the MI355X analog of the reference's in-pod
+``nvidia-smi`` payload, /root/reference/nvidia-smi.yaml:13).
+"""
+
+from __future__ import annotations
+
+_C = None
+_import_error: Exception | None = None
+
+try:
+    import torch  # noqa: F401  — the extension links against torch's libs
+    from k3samd import _C  # type: ignore[attr-defined]
+except ImportError as e:  # extension not built (CPU-only dev box is fine)
+    _import_error = e
+
+
+def _require_native():
+    if _C is None:
+        raise RuntimeError(
+            "k3samd native extension is not built. Run "
+            "`python -m k3samd.build` (or __graft_entry__.build()). "
+            f"Original import error: {_import_error}"
+        )
+    return _C
+
+
+def native_available() -> bool:
+    return _C is not None
+
+
+def stream_triad(a, b, c, s, nontemporal=False):
+    """a = b + s*c over fp32 tensors, 16B/lane vectorized CDNA4 kernel."""
+    _require_native().stream_triad(a, b, c, float(s), nontemporal)
+
+
+def stream_copy(a, b, nontemporal=False):
+    _require_native().stream_copy(a, b, nontemporal)
+
+
+def stream_scale(a, c, s, nontemporal=False):
+    _require_native().stream_scale(a, c, float(s), nontemporal)
+
+
+def stream_add(a, b, c, nontemporal=False):
+    _require_native().stream_add(a, b, c, nontemporal)
+
+
+def mfma_throughput(out, iters):
+    """Issue `iters` MFMA quads per wave; returns FLOPs issued."""
+    return _require_native().mfma_throughput(out, int(iters))
+
+
+def mfma_gemm16(A, B, layout=0):
+    """Single-tile D[16,16] = A[16,32] @ B[32,16] via one bf16 MFMA."""
+    return _require_native().mfma_gemm16(A, B, int(layout))
