@@ -55,6 +55,13 @@ def mfma_throughput(out, iters):
     return _require_native().mfma_throughput(out, int(iters))
 
 
-def mfma_gemm16(A, B, layout=0):
+# Validated on MI355X silicon (first gpurun, 2026-09-13): layout 0 — lane l
+# holds A[l&15][(l>>4)*8 + j] (k-contiguous 8-element groups) — matches the
+# hardware fragment mapping of v_mfma_f32_16x16x32_bf16 (max err 1.9e-6 vs
+# torch fp32 matmul); layout 1 does not.
+MFMA_LAYOUT = 0
+
+
+def mfma_gemm16(A, B, layout=MFMA_LAYOUT):
     """Single-tile D[16,16] = A[16,32] @ B[32,16] via one bf16 MFMA."""
     return _require_native().mfma_gemm16(A, B, int(layout))

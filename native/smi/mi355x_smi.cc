@@ -1,0 +1,144 @@
+// mi355x-smi — GPU status CLI for the smoke pod.
+//
+// The MI355X-native analog of the `nvidia-smi` payload the reference runs in
+// its validation pod (/root/reference/nvidia-smi.yaml:13) and of the
+// golden-output blocks the reference's docs assert on
+// (/root/reference/README.md:71-93, :137-156). Reads the KFD sysfs topology
+// directly (native/topology), plus per-card DRM runtime stats where present.
+//
+//   mi355x-smi            human-readable table
+//   mi355x-smi --json     machine-readable (used by tests/fixtures)
+//
+// Honors K3SAMD_SYSFS_ROOT for fixture-tree testing on CPU-only boxes.
+
+#include <cstdio>
+#include <cstring>
+#include <fstream>
+#include <sstream>
+#include <string>
+
+#include "../common/json_writer.h"
+#include "../topology/kfd_topology.h"
+
+namespace {
+
+std::string read_trim(const std::string& path) {
+  std::ifstream f(path);
+  if (!f) return {};
+  std::stringstream ss;
+  ss << f.rdbuf();
+  std::string s = ss.str();
+  while (!s.empty() && (s.back() == '\n' || s.back() == ' ')) s.pop_back();
+  return s;
+}
+
+uint64_t read_u64(const std::string& path, uint64_t dflt = 0) {
+  std::string s = read_trim(path);
+  if (s.empty()) return dflt;
+  return std::strtoull(s.c_str(), nullptr, 10);
+}
+
+struct CardStats {
+  uint64_t vram_used = 0;
+  uint64_t vram_total = 0;
+  long busy_percent = -1;
+  long temp_mc = -1;  // millidegrees C
+};
+
+CardStats card_stats(const std::string& sysfs_root, int card_index) {
+  CardStats st;
+  if (card_index < 0) return st;
+  std::string dev = sysfs_root + "/class/drm/card" +
+                    std::to_string(card_index) + "/device";
+  st.vram_used = read_u64(dev + "/mem_info_vram_used");
+  st.vram_total = read_u64(dev + "/mem_info_vram_total");
+  std::string busy = read_trim(dev + "/gpu_busy_percent");
+  if (!busy.empty()) st.busy_percent = std::strtol(busy.c_str(), nullptr, 10);
+  // first hwmon temp input if present
+  for (int h = 0; h < 8; ++h) {
+    std::string t = read_trim(dev + "/hwmon/hwmon" + std::to_string(h) +
+                              "/temp1_input");
+    if (!t.empty()) {
+      st.temp_mc = std::strtol(t.c_str(), nullptr, 10);
+      break;
+    }
+  }
+  return st;
+}
+
+double mib(uint64_t b) { return (double)b / (1024.0 * 1024.0); }
+
+}  // namespace
+
+int main(int argc, char** argv) {
+  bool json = false;
+  for (int i = 1; i < argc; ++i) {
+    if (!std::strcmp(argv[i], "--json")) json = true;
+    else if (!std::strcmp(argv[i], "--help") || !std::strcmp(argv[i], "-h")) {
+      std::printf("usage: mi355x-smi [--json]\n");
+      return 0;
+    }
+  }
+
+  const std::string root = k3samd::default_sysfs_root();
+  k3samd::Topology topo = k3samd::enumerate_topology(root);
+
+  if (json) {
+    k3samd::JsonWriter w;
+    w.begin_obj();
+    w.key("driver_version").value(topo.driver_version);
+    w.key("gpu_count").value((uint64_t)topo.gpus.size());
+    w.key("gpus").begin_arr();
+    for (const auto& g : topo.gpus) {
+      CardStats st = card_stats(root, g.card_index);
+      w.begin_obj();
+      w.key("kfd_node").value(g.kfd_node);
+      w.key("id").value(g.stable_id());
+      w.key("name").value(g.name);
+      w.key("arch").value(g.gfx_arch());
+      w.key("pci_bdf").value(g.pci_bdf);
+      w.key("render_minor").value(g.drm_render_minor);
+      w.key("card_index").value(g.card_index);
+      w.key("vram_bytes").value(g.vram_bytes);
+      w.key("vram_used_bytes").value(st.vram_used);
+      w.key("compute_units").value(g.compute_units());
+      w.key("xgmi_links").value(g.xgmi_links);
+      w.key("numa_node").value(g.numa_node);
+      w.key("busy_percent").value((int64_t)st.busy_percent);
+      w.key("temp_milli_c").value((int64_t)st.temp_mc);
+      w.end_obj();
+    }
+    w.end_arr();
+    w.end_obj();
+    std::printf("%s\n", w.str().c_str());
+    return 0;
+  }
+
+  std::printf("+---------------------------------------------------------------------------+\n");
+  std::printf("| mi355x-smi            driver: %-12s                 k3samd stack   |\n",
+              topo.driver_version.empty() ? "unknown" : topo.driver_version.c_str());
+  std::printf("+----+----------------------+--------+-----+-------------------+------+-----+\n");
+  std::printf("| ## | Name                 | Arch   | CUs | VRAM used / total | xGMI | Tmp |\n");
+  std::printf("+----+----------------------+--------+-----+-------------------+------+-----+\n");
+  int i = 0;
+  for (const auto& g : topo.gpus) {
+    CardStats st = card_stats(root, g.card_index);
+    char vram[40];
+    std::snprintf(vram, sizeof(vram), "%6.0f/%6.0f MiB", mib(st.vram_used),
+                  mib(g.vram_bytes));
+    char temp[16];
+    if (st.temp_mc >= 0)
+      std::snprintf(temp, sizeof(temp), "%2ldC", st.temp_mc / 1000);
+    else
+      std::snprintf(temp, sizeof(temp), " - ");
+    std::printf("| %2d | %-20.20s | %-6s | %3u | %-17s | %4d | %s |\n", i++,
+                g.name.c_str(), g.gfx_arch().c_str(), g.compute_units(), vram,
+                g.xgmi_links, temp);
+  }
+  if (topo.gpus.empty()) {
+    std::printf("| no AMD GPUs found (no KFD topology under %s)\n",
+                root.c_str());
+  }
+  std::printf("+----+----------------------+--------+-----+-------------------+------+-----+\n");
+  return 0;
+}
