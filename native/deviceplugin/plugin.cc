@@ -1,0 +1,331 @@
+#include "plugin.h"
+
+#include <algorithm>
+#include <chrono>
+#include <cstdio>
+#include <set>
+
+#include "../common/miniyaml.h"
+
+namespace k3samd {
+
+bool PluginConfig::from_yaml(const std::string& text, PluginConfig& out,
+                             std::string* err) {
+  try {
+    YNode root = yaml_parse(text);
+    if (const YNode* v = root.get("version")) out.version = v->as_str("v1");
+    if (out.version != "v1") {
+      if (err) *err = "unsupported config version: " + out.version;
+      return false;
+    }
+    if (const YNode* m = root.get_path("flags.migStrategy"))
+      out.mig_strategy = m->as_str("none");
+    if (out.mig_strategy != "none") {
+      // MI355X partitioning (SR-IOV/NPS) is not scheduling-level sharing;
+      // mirror the reference's `migStrategy: none` (values.yaml:11) only.
+      if (err) *err = "only migStrategy 'none' is supported";
+      return false;
+    }
+    const YNode* ts = root.get_path("sharing.timeSlicing");
+    if (ts) {
+      if (const YNode* n = ts->get("renameByDefault"))
+        out.rename_by_default = n->as_bool(false);
+      if (const YNode* n = ts->get("failRequestsGreaterThanOne"))
+        out.fail_requests_greater_than_one = n->as_bool(false);
+      if (const YNode* rs = ts->get("resources")) {
+        for (const YNode& r : rs->list) {
+          std::string name =
+              r.get("name") ? r.get("name")->as_str() : out.resource_name;
+          if (name == out.resource_name) {
+            if (const YNode* rep = r.get("replicas")) {
+              out.replicas = (int)rep->as_int(1);
+              if (out.replicas < 1 || out.replicas > 256) {
+                if (err) *err = "replicas out of range";
+                return false;
+              }
+            }
+          }
+        }
+      }
+    }
+    return true;
+  } catch (const std::exception& e) {
+    if (err) *err = e.what();
+    return false;
+  }
+}
+
+DevicePlugin::DevicePlugin(PluginConfig cfg, std::string sysfs_root,
+                           std::string dev_root)
+    : cfg_(std::move(cfg)),
+      sysfs_root_(std::move(sysfs_root)),
+      dev_root_(std::move(dev_root)) {
+  refresh_topology();
+}
+
+DevicePlugin::~DevicePlugin() { stop(); }
+
+void DevicePlugin::refresh_topology() {
+  Topology topo = enumerate_topology(sysfs_root_);
+  std::lock_guard<std::mutex> lk(mu_);
+  topo_ = std::move(topo);
+  devices_.clear();
+  for (size_t gi = 0; gi < topo_.gpus.size(); ++gi) {
+    const auto& g = topo_.gpus[gi];
+    if (cfg_.replicas <= 1) {
+      devices_.push_back({g.stable_id(), (int)gi, true});
+    } else {
+      for (int r = 0; r < cfg_.replicas; ++r) {
+        devices_.push_back(
+            {g.stable_id() + "::" + std::to_string(r), (int)gi, true});
+      }
+    }
+  }
+  ++generation_;
+  gen_cv_.notify_all();
+}
+
+std::vector<VirtualDevice> DevicePlugin::devices() const {
+  std::lock_guard<std::mutex> lk(mu_);
+  return devices_;
+}
+
+std::string DevicePlugin::advertised_resource() const {
+  // NVIDIA-plugin semantics (README.md:112 behavior surface): with sharing
+  // enabled and renameByDefault, the resource is advertised as
+  // "<name>.shared".
+  if (cfg_.replicas > 1 && cfg_.rename_by_default)
+    return cfg_.resource_name + ".shared";
+  return cfg_.resource_name;
+}
+
+std::string DevicePlugin::handle_options() {
+  dp::DevicePluginOptions opts;
+  opts.get_preferred_allocation_available = true;
+  opts.pre_start_required = false;
+  return opts.encode();
+}
+
+std::vector<dp::Device> DevicePlugin::current_device_list() {
+  std::lock_guard<std::mutex> lk(mu_);
+  std::vector<dp::Device> out;
+  for (const auto& vd : devices_) {
+    dp::Device d;
+    d.id = vd.id;
+    d.health = vd.healthy ? dp::kHealthy : dp::kUnhealthy;
+    d.numa_node = topo_.gpus[vd.gpu_index].numa_node;
+    out.push_back(std::move(d));
+  }
+  return out;
+}
+
+int DevicePlugin::gpu_for_id(const std::string& vid) const {
+  for (const auto& vd : devices_)
+    if (vd.id == vid) return vd.gpu_index;
+  return -1;
+}
+
+GrpcStatus DevicePlugin::handle_allocate(const std::string& req,
+                                         std::string& resp) {
+  std::vector<std::vector<std::string>> containers;
+  if (!dp::decode_allocate_request(req, containers))
+    return {13, "malformed AllocateRequest"};
+
+  std::lock_guard<std::mutex> lk(mu_);
+  std::vector<dp::ContainerAllocateResponse> crs;
+  for (const auto& ids : containers) {
+    if (cfg_.replicas > 1 && cfg_.fail_requests_greater_than_one &&
+        ids.size() > 1) {
+      return {3,
+              "request for more than one " + advertised_resource() +
+                  " is not allowed with time-slicing "
+                  "(failRequestsGreaterThanOne=true)"};
+    }
+    dp::ContainerAllocateResponse cr;
+    std::set<int> gpus;  // dedupe physical GPUs across replica ids
+    for (const auto& id : ids) {
+      int gi = -1;
+      for (const auto& vd : devices_)
+        if (vd.id == id) {
+          gi = vd.gpu_index;
+          break;
+        }
+      if (gi < 0) return {3, "unknown device id " + id};
+      gpus.insert(gi);
+    }
+    // /dev/kfd is the compute entry point, shared by all GPUs
+    cr.devices.push_back({"/dev/kfd", dev_root_ + "/kfd", "rw"});
+    std::string visible, minors;
+    for (int gi : gpus) {
+      const auto& g = topo_.gpus[gi];
+      std::string rnode =
+          "/dri/renderD" + std::to_string(g.drm_render_minor);
+      cr.devices.push_back({"/dev" + rnode, dev_root_ + rnode, "rw"});
+      if (g.card_index >= 0) {
+        std::string cnode = "/dri/card" + std::to_string(g.card_index);
+        cr.devices.push_back({"/dev" + cnode, dev_root_ + cnode, "rw"});
+      }
+      if (!visible.empty()) visible += ",";
+      visible += g.stable_id();
+      if (!minors.empty()) minors += ",";
+      minors += std::to_string(g.drm_render_minor);
+    }
+    cr.envs["K3SAMD_VISIBLE_DEVICES"] = visible;
+    cr.envs["K3SAMD_RENDER_MINORS"] = minors;
+    cr.annotations["k3samd.ai/allocated-gpus"] = visible;
+    crs.push_back(std::move(cr));
+  }
+  resp = dp::encode_allocate_response(crs);
+  return GrpcStatus::Ok();
+}
+
+GrpcStatus DevicePlugin::handle_preferred(const std::string& req,
+                                          std::string& resp) {
+  std::vector<dp::PreferredRequest> reqs;
+  if (!dp::decode_preferred_request(req, reqs))
+    return {13, "malformed PreferredAllocationRequest"};
+
+  std::lock_guard<std::mutex> lk(mu_);
+  std::vector<std::vector<std::string>> out;
+  for (const auto& pr : reqs) {
+    std::vector<std::string> chosen(pr.must_include);
+    std::set<std::string> used(chosen.begin(), chosen.end());
+
+    // Prefer packing replicas of the same physical GPU together (keeps
+    // other physical GPUs free for exclusive jobs), then same-NUMA spread.
+    std::vector<std::string> avail;
+    for (const auto& id : pr.available)
+      if (!used.count(id)) avail.push_back(id);
+    std::stable_sort(avail.begin(), avail.end(),
+                     [&](const std::string& a, const std::string& b) {
+                       int ga = gpu_for_id(a), gb = gpu_for_id(b);
+                       if (ga != gb) return ga < gb;
+                       return a < b;
+                     });
+    for (const auto& id : avail) {
+      if ((int)chosen.size() >= pr.size) break;
+      chosen.push_back(id);
+    }
+    if ((int)chosen.size() > pr.size) chosen.resize(pr.size);
+    out.push_back(std::move(chosen));
+  }
+  resp = dp::encode_preferred_response(out);
+  return GrpcStatus::Ok();
+}
+
+bool DevicePlugin::poll_health_once() {
+  Topology topo = enumerate_topology(sysfs_root_);
+  std::set<std::string> present;
+  for (const auto& g : topo.gpus) present.insert(g.stable_id());
+
+  std::lock_guard<std::mutex> lk(mu_);
+  bool changed = false;
+  for (auto& vd : devices_) {
+    const auto& g = topo_.gpus[vd.gpu_index];
+    bool healthy = present.count(g.stable_id()) > 0;
+    if (healthy != vd.healthy) {
+      vd.healthy = healthy;
+      changed = true;
+    }
+  }
+  if (changed) {
+    ++generation_;
+    gen_cv_.notify_all();
+  }
+  return changed;
+}
+
+GrpcStatus DevicePlugin::handle_list_and_watch(
+    const std::string&, const GrpcServer::WriteFn& write) {
+  uint64_t seen_gen;
+  {
+    std::lock_guard<std::mutex> lk(mu_);
+    seen_gen = generation_;
+  }
+  if (!write(dp::encode_list_and_watch(current_device_list())))
+    return GrpcStatus::Ok();
+  // push an update whenever the device list changes (kubelet keeps this
+  // stream open for the plugin's lifetime)
+  while (!stopping_.load()) {
+    std::unique_lock<std::mutex> lk(mu_);
+    gen_cv_.wait_for(lk, std::chrono::seconds(1),
+                     [&] { return generation_ != seen_gen || stopping_.load(); });
+    if (stopping_.load()) break;
+    if (generation_ == seen_gen) continue;
+    seen_gen = generation_;
+    lk.unlock();
+    if (!write(dp::encode_list_and_watch(current_device_list()))) break;
+  }
+  return GrpcStatus::Ok();
+}
+
+bool DevicePlugin::serve(const std::string& plugin_sock,
+                         const std::string& kubelet_sock, int health_poll_ms) {
+  server_.add_unary(dp::kOptionsPath,
+                    [this](const std::string&, std::string& resp) {
+                      resp = handle_options();
+                      return GrpcStatus::Ok();
+                    });
+  server_.add_unary(dp::kAllocatePath,
+                    [this](const std::string& req, std::string& resp) {
+                      return handle_allocate(req, resp);
+                    });
+  server_.add_unary(dp::kPreferredPath,
+                    [this](const std::string& req, std::string& resp) {
+                      return handle_preferred(req, resp);
+                    });
+  server_.add_unary(dp::kPreStartPath,
+                    [](const std::string&, std::string& resp) {
+                      resp.clear();
+                      return GrpcStatus::Ok();
+                    });
+  server_.add_server_stream(
+      dp::kListAndWatchPath,
+      [this](const std::string& req, const GrpcServer::WriteFn& write) {
+        return handle_list_and_watch(req, write);
+      });
+
+  if (!server_.start(plugin_sock)) {
+    std::fprintf(stderr, "deviceplugin: cannot bind %s\n",
+                 plugin_sock.c_str());
+    return false;
+  }
+
+  if (!kubelet_sock.empty()) {
+    dp::RegisterRequest rr;
+    size_t slash = plugin_sock.find_last_of('/');
+    rr.endpoint =
+        slash == std::string::npos ? plugin_sock : plugin_sock.substr(slash + 1);
+    rr.resource_name = advertised_resource();
+    auto res = grpc_unary_call(kubelet_sock, dp::kRegisterPath, rr.encode());
+    if (!res.transport_ok || res.grpc_status != 0) {
+      std::fprintf(stderr,
+                   "deviceplugin: Register with kubelet failed: %s (grpc=%d %s)\n",
+                   res.error.c_str(), res.grpc_status,
+                   res.grpc_message.c_str());
+      server_.stop();
+      return false;
+    }
+  }
+
+  stopping_.store(false);
+  if (health_poll_ms > 0) {
+    health_thread_ = std::thread([this, health_poll_ms] {
+      while (!stopping_.load()) {
+        std::this_thread::sleep_for(std::chrono::milliseconds(health_poll_ms));
+        if (stopping_.load()) break;
+        poll_health_once();
+      }
+    });
+  }
+  return true;
+}
+
+void DevicePlugin::stop() {
+  stopping_.store(true);
+  gen_cv_.notify_all();
+  if (health_thread_.joinable()) health_thread_.join();
+  server_.stop();
+}
+
+}  // namespace k3samd

@@ -1,0 +1,293 @@
+"""gRPC interop tests: the C++ device plugin against grpcio (the same gRPC
+core family kubelet's grpc-go belongs to).
+
+Covers the reference stack's scheduling surface end-to-end without a
+cluster: registration handshake, ListAndWatch device
+advertisement with time-slice replica fan-out (values.yaml:12-18 semantics),
+Allocate device injection, preferred allocation, failRequestsGreaterThanOne,
+and health transitions pushed over the live ListAndWatch stream.
+"""
+
+import json
+import shutil
+import subprocess
+import threading
+import time
+from concurrent import futures
+from pathlib import Path
+
+import grpc
+import pytest
+
+import pb_v1beta1 as pb
+from sysfs_builder import build_tree
+
+REPO = Path(__file__).resolve().parent.parent
+PLUGIN = REPO / "native" / "bin" / "k3samd-device-plugin"
+
+IDENT = lambda b: b  # noqa: E731  identity (de)serializer
+
+
+@pytest.fixture(scope="session", autouse=True)
+def build_native():
+    subprocess.run(["make", "-C", str(REPO / "native"), "-j8"], check=True,
+                   capture_output=True)
+
+
+class FakeKubelet:
+    """Records Register calls, like kubelet's Registration service."""
+
+    def __init__(self, sock_path):
+        self.requests = []
+        self.event = threading.Event()
+        handler = grpc.method_handlers_generic_handler(
+            "v1beta1.Registration",
+            {"Register": grpc.unary_unary_rpc_method_handler(
+                self._register,
+                request_deserializer=IDENT, response_serializer=IDENT)})
+        self.server = grpc.server(futures.ThreadPoolExecutor(max_workers=2))
+        self.server.add_generic_rpc_handlers((handler,))
+        self.server.add_insecure_port(f"unix:{sock_path}")
+        self.server.start()
+
+    def _register(self, request, context):
+        self.requests.append(pb.decode_register_request(request))
+        self.event.set()
+        return b""
+
+    def stop(self):
+        self.server.stop(0)
+
+
+DEFAULT_CFG = """\
+version: v1
+flags:
+  migStrategy: none
+sharing:
+  timeSlicing:
+    renameByDefault: false
+    failRequestsGreaterThanOne: false
+    resources:
+    - name: amd.com/gpu
+      replicas: {replicas}
+"""
+
+
+class PluginHarness:
+    def __init__(self, tmp_path, n_gpus=8, replicas=1, cfg_text=None,
+                 register=True, health_poll_ms=200):
+        self.root = build_tree(tmp_path / "sys", n_gpus=n_gpus)
+        self.dir = tmp_path
+        cfg = tmp_path / "config.yaml"
+        cfg.write_text(cfg_text or DEFAULT_CFG.format(replicas=replicas))
+        self.plugin_sock = str(tmp_path / "amd-gpu.sock")
+        self.kubelet_sock = str(tmp_path / "kubelet.sock")
+        self.kubelet = FakeKubelet(self.kubelet_sock) if register else None
+        argv = [str(PLUGIN), "--config", str(cfg),
+                "--plugin-sock", self.plugin_sock,
+                "--health-poll-ms", str(health_poll_ms)]
+        if register:
+            argv += ["--kubelet-sock", self.kubelet_sock]
+        else:
+            argv += ["--no-register"]
+        self.proc = subprocess.Popen(
+            argv, env={"K3SAMD_SYSFS_ROOT": str(self.root)},
+            stdout=subprocess.PIPE, stderr=subprocess.PIPE)
+        deadline = time.time() + 10
+        while not Path(self.plugin_sock).exists():
+            if time.time() > deadline or self.proc.poll() is not None:
+                raise RuntimeError(
+                    f"plugin did not start: {self.proc.stderr.read()}")
+            time.sleep(0.02)
+        self.channel = grpc.insecure_channel(f"unix:{self.plugin_sock}")
+
+    def call(self, method, request=b"", timeout=5):
+        fn = self.channel.unary_unary(f"/v1beta1.DevicePlugin/{method}",
+                                      request_serializer=IDENT,
+                                      response_deserializer=IDENT)
+        return fn(request, timeout=timeout)
+
+    def stream(self, method, request=b"", timeout=10):
+        fn = self.channel.unary_stream(f"/v1beta1.DevicePlugin/{method}",
+                                       request_serializer=IDENT,
+                                       response_deserializer=IDENT)
+        return fn(request, timeout=timeout)
+
+    def close(self):
+        self.channel.close()
+        self.proc.terminate()
+        try:
+            self.proc.wait(timeout=5)
+        except subprocess.TimeoutExpired:
+            self.proc.kill()
+        if self.kubelet:
+            self.kubelet.stop()
+
+
+@pytest.fixture
+def harness(tmp_path):
+    hs = []
+
+    def make(**kw):
+        h = PluginHarness(tmp_path, **kw)
+        hs.append(h)
+        return h
+
+    yield make
+    for h in hs:
+        h.close()
+
+
+def test_register_handshake(harness):
+    h = harness(n_gpus=8, replicas=1)
+    assert h.kubelet.event.wait(5), "plugin never registered"
+    req = h.kubelet.requests[0]
+    assert req["version"] == "v1beta1"
+    assert req["endpoint"] == "amd-gpu.sock"
+    assert req["resource_name"] == "amd.com/gpu"
+    assert req["options"]["get_preferred_allocation_available"] is True
+
+
+def test_options(harness):
+    h = harness(n_gpus=2, register=False)
+    opts = pb.decode_options(h.call("GetDevicePluginOptions"))
+    assert opts["get_preferred_allocation_available"] is True
+    assert opts["pre_start_required"] is False
+
+
+def test_list_and_watch_exclusive(harness):
+    h = harness(n_gpus=8, replicas=1, register=False)
+    stream = h.stream("ListAndWatch")
+    devs = pb.decode_list_and_watch(next(stream))
+    assert len(devs) == 8
+    assert all(d["health"] == "Healthy" for d in devs)
+    assert all(d["id"].startswith("amdgpu-") for d in devs)
+    assert {d["numa"] for d in devs} == {0, 1}  # fixture alternates sockets
+
+
+def test_list_and_watch_timesliced(harness):
+    """values.yaml:17-18 semantics: replicas=4 => 4 virtual devices per GPU
+    (32 allocatable on the 8-GPU node, BASELINE.md structural target)."""
+    h = harness(n_gpus=8, replicas=4, register=False)
+    devs = pb.decode_list_and_watch(next(h.stream("ListAndWatch")))
+    assert len(devs) == 32
+    suffixes = [d["id"].rsplit("::", 1)[1] for d in devs]
+    assert suffixes.count("0") == 8 and suffixes.count("3") == 8
+
+
+def test_allocate_single(harness):
+    h = harness(n_gpus=8, replicas=1, register=False)
+    devs = pb.decode_list_and_watch(next(h.stream("ListAndWatch")))
+    resp = pb.decode_allocate_response(
+        h.call("Allocate", pb.encode_allocate_request([[devs[0]["id"]]])))
+    assert len(resp) == 1
+    cr = resp[0]
+    paths = {d["host_path"] for d in cr["devices"]}
+    assert "/dev/kfd" in paths
+    assert "/dev/dri/renderD128" in paths
+    assert "/dev/dri/card0" in paths
+    assert cr["envs"]["K3SAMD_VISIBLE_DEVICES"] == devs[0]["id"]
+    assert cr["envs"]["K3SAMD_RENDER_MINORS"] == "128"
+    assert all(d["permissions"] == "rw" for d in cr["devices"])
+
+
+def test_allocate_all_eight(harness):
+    h = harness(n_gpus=8, replicas=1, register=False)
+    devs = pb.decode_list_and_watch(next(h.stream("ListAndWatch")))
+    ids = [d["id"] for d in devs]
+    resp = pb.decode_allocate_response(
+        h.call("Allocate", pb.encode_allocate_request([ids])))
+    cr = resp[0]
+    assert len(cr["devices"]) == 1 + 8 + 8  # kfd + 8 render + 8 card
+    minors = cr["envs"]["K3SAMD_RENDER_MINORS"].split(",")
+    assert sorted(minors) == sorted(str(m) for m in range(128, 136))
+
+
+def test_allocate_replicas_dedupe(harness):
+    """Two replicas of the same physical GPU inject that GPU once."""
+    h = harness(n_gpus=2, replicas=4, register=False)
+    devs = pb.decode_list_and_watch(next(h.stream("ListAndWatch")))
+    base = devs[0]["id"].rsplit("::", 1)[0]
+    ids = [f"{base}::0", f"{base}::1"]
+    resp = pb.decode_allocate_response(
+        h.call("Allocate", pb.encode_allocate_request([ids])))
+    cr = resp[0]
+    render = [d for d in cr["devices"] if "renderD" in d["host_path"]]
+    assert len(render) == 1
+
+
+def test_allocate_unknown_id(harness):
+    h = harness(n_gpus=1, register=False)
+    with pytest.raises(grpc.RpcError) as ei:
+        h.call("Allocate", pb.encode_allocate_request([["nope"]]))
+    assert ei.value.code() == grpc.StatusCode.INVALID_ARGUMENT
+
+
+def test_fail_requests_greater_than_one(harness):
+    cfg = DEFAULT_CFG.format(replicas=4).replace(
+        "failRequestsGreaterThanOne: false",
+        "failRequestsGreaterThanOne: true")
+    h = harness(n_gpus=2, cfg_text=cfg, register=False)
+    devs = pb.decode_list_and_watch(next(h.stream("ListAndWatch")))
+    ids = [devs[0]["id"], devs[1]["id"]]
+    with pytest.raises(grpc.RpcError) as ei:
+        h.call("Allocate", pb.encode_allocate_request([ids]))
+    assert ei.value.code() == grpc.StatusCode.INVALID_ARGUMENT
+    # single-device requests still succeed
+    resp = pb.decode_allocate_response(
+        h.call("Allocate", pb.encode_allocate_request([[devs[0]["id"]]])))
+    assert resp[0]["devices"]
+
+
+def test_rename_by_default(harness):
+    cfg = DEFAULT_CFG.format(replicas=4).replace(
+        "renameByDefault: false", "renameByDefault: true")
+    h = harness(n_gpus=1, cfg_text=cfg)
+    assert h.kubelet.event.wait(5)
+    assert h.kubelet.requests[0]["resource_name"] == "amd.com/gpu.shared"
+
+
+def test_preferred_allocation_packs_replicas(harness):
+    h = harness(n_gpus=2, replicas=4, register=False)
+    devs = pb.decode_list_and_watch(next(h.stream("ListAndWatch")))
+    avail = [d["id"] for d in devs]
+    resp = pb.decode_preferred_response(
+        h.call("GetPreferredAllocation",
+               pb.encode_preferred_request(avail, [], 3)))
+    assert len(resp) == 1 and len(resp[0]) == 3
+    # all three picks should share one physical GPU
+    bases = {i.rsplit("::", 1)[0] for i in resp[0]}
+    assert len(bases) == 1
+
+
+def test_health_transition_pushes_update(harness):
+    h = harness(n_gpus=2, replicas=1, register=False, health_poll_ms=100)
+    stream = h.stream("ListAndWatch", timeout=30)
+    first = pb.decode_list_and_watch(next(stream))
+    assert all(d["health"] == "Healthy" for d in first)
+    # GPU 1 dies: its KFD node vanishes from sysfs
+    shutil.rmtree(h.root / "class/kfd/kfd/topology/nodes/3")
+    second = pb.decode_list_and_watch(next(stream))
+    by_health = sorted(d["health"] for d in second)
+    assert by_health == ["Healthy", "Unhealthy"]
+
+
+def test_cpu_only_zero_allocatable(tmp_path):
+    """BASELINE.json config #1: CPU-only node => 0 amd.com/gpu."""
+    root = build_tree(tmp_path / "sys", n_gpus=0)
+    out = subprocess.run(
+        [str(PLUGIN), "--oneshot"],
+        env={"K3SAMD_SYSFS_ROOT": str(root)},
+        capture_output=True, text=True, timeout=60)
+    j = json.loads(out.stdout)
+    assert j["resource"] == "amd.com/gpu"
+    assert j["allocatable"] == 0
+
+
+def test_bad_config_rejected(tmp_path):
+    cfg = tmp_path / "bad.yaml"
+    cfg.write_text("version: v2\n")
+    out = subprocess.run([str(PLUGIN), "--config", str(cfg), "--oneshot"],
+                         capture_output=True, text=True, timeout=60)
+    assert out.returncode == 2
+    assert "version" in out.stderr
