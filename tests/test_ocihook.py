@@ -1,0 +1,160 @@
+"""Golden config.json tests for the OCI runtime wrapper.
+
+Mirrors the reference's container-runtime layer contract
+(/root/reference/README.md:57-69, :164): a pod with the GPU RuntimeClass
+gets device nodes + cgroup rules (+ optionally ROCm userspace) injected,
+scoped to the devices the device plugin allocated.
+"""
+
+import json
+import subprocess
+from pathlib import Path
+
+import pytest
+
+from sysfs_builder import build_tree
+
+REPO = Path(__file__).resolve().parent.parent
+RUNTIME = REPO / "native" / "bin" / "k3samd-oci-runtime"
+
+
+@pytest.fixture(scope="session", autouse=True)
+def build_native():
+    subprocess.run(["make", "-C", str(REPO / "native"), "-j8"], check=True,
+                   capture_output=True)
+
+
+BASE_SPEC = {
+    "ociVersion": "1.0.2",
+    "process": {
+        "args": ["rocm-smi"],
+        "env": ["PATH=/usr/bin"],
+        "cwd": "/",
+    },
+    "root": {"path": "rootfs"},
+    "mounts": [
+        {"destination": "/proc", "type": "proc", "source": "proc"},
+    ],
+    "linux": {
+        "namespaces": [{"type": "pid"}, {"type": "mount"}],
+        "resources": {"devices": [{"allow": False, "access": "rwm"}]},
+    },
+}
+
+
+def transform(tmp_path, spec, n_gpus=8, extra_env=None):
+    sysfs = build_tree(tmp_path / "sys", n_gpus=n_gpus)
+    cfg = tmp_path / "config.json"
+    spec = json.loads(json.dumps(spec))  # deep copy
+    if extra_env:
+        spec["process"]["env"].extend(extra_env)
+    cfg.write_text(json.dumps(spec))
+    proc = subprocess.run(
+        [str(RUNTIME), "--transform-only", str(cfg)],
+        env={"K3SAMD_SYSFS_ROOT": str(sysfs),
+             "K3SAMD_DEV_ROOT": str(tmp_path / "nonexistent-dev")},
+        capture_output=True, text=True, timeout=60)
+    assert proc.returncode == 0, proc.stderr
+    return json.loads(cfg.read_text()), proc.stderr
+
+
+def device_paths(spec):
+    return [d["path"] for d in spec["linux"]["devices"]]
+
+
+def test_default_injects_all_gpus(tmp_path):
+    out, _ = transform(tmp_path, BASE_SPEC, n_gpus=8)
+    paths = device_paths(out)
+    assert "/dev/kfd" in paths
+    for k in range(8):
+        assert f"/dev/dri/renderD{128 + k}" in paths
+        assert f"/dev/dri/card{k}" in paths
+    assert len(paths) == 17
+    # cgroup allow rules added alongside, preserving the existing deny rule
+    rules = out["linux"]["resources"]["devices"]
+    assert rules[0] == {"allow": False, "access": "rwm"}
+    allows = [r for r in rules if r.get("allow")]
+    assert len(allows) == 17
+    for r in allows:
+        assert r["type"] == "c" and r["access"] == "rwm"
+
+
+def test_visible_devices_subset(tmp_path):
+    env = ["K3SAMD_VISIBLE_DEVICES=amdgpu-1a2b3c4d5e6f0002",
+           "K3SAMD_RENDER_MINORS=130"]
+    out, _ = transform(tmp_path, BASE_SPEC, n_gpus=8, extra_env=env)
+    paths = device_paths(out)
+    assert paths == ["/dev/kfd", "/dev/dri/renderD130", "/dev/dri/card2"]
+
+
+def test_visible_none_skips(tmp_path):
+    out, err = transform(tmp_path, BASE_SPEC, n_gpus=8,
+                         extra_env=["K3SAMD_VISIBLE_DEVICES=none"])
+    assert "devices" not in out["linux"] or not any(
+        "kfd" in p for p in device_paths(out))
+    assert "skipped" in err
+
+
+def test_unrelated_fields_roundtrip(tmp_path):
+    spec = json.loads(json.dumps(BASE_SPEC))
+    spec["annotations"] = {"io.kubernetes.pod.name": "rocm-smi"}
+    spec["hooks"] = {"prestart": [{"path": "/bin/true"}]}
+    spec["linux"]["seccomp"] = {"defaultAction": "SCMP_ACT_ALLOW"}
+    out, _ = transform(tmp_path, spec, n_gpus=1)
+    assert out["annotations"] == spec["annotations"]
+    assert out["hooks"] == spec["hooks"]
+    assert out["linux"]["seccomp"] == spec["linux"]["seccomp"]
+    assert out["mounts"][0]["destination"] == "/proc"
+    assert out["process"]["args"] == ["rocm-smi"]
+
+
+def test_idempotent(tmp_path):
+    sysfs = build_tree(tmp_path / "sys", n_gpus=2)
+    cfg = tmp_path / "config.json"
+    cfg.write_text(json.dumps(BASE_SPEC))
+    env = {"K3SAMD_SYSFS_ROOT": str(sysfs),
+           "K3SAMD_DEV_ROOT": str(tmp_path / "nodev")}
+    for _ in range(2):
+        subprocess.run([str(RUNTIME), "--transform-only", str(cfg)],
+                       env=env, check=True, capture_output=True, timeout=60)
+    out = json.loads(cfg.read_text())
+    paths = device_paths(out)
+    assert len(paths) == len(set(paths)) == 5  # kfd + 2x(render+card)
+    allows = [r for r in out["linux"]["resources"]["devices"]
+              if r.get("allow")]
+    assert len(allows) == 5
+
+
+def test_rocm_mount_injection(tmp_path):
+    rocm = tmp_path / "opt-rocm"
+    rocm.mkdir()
+    sysfs = build_tree(tmp_path / "sys", n_gpus=1)
+    cfg = tmp_path / "config.json"
+    spec = json.loads(json.dumps(BASE_SPEC))
+    spec["process"]["env"].append("K3SAMD_INJECT_ROCM=1")
+    cfg.write_text(json.dumps(spec))
+    subprocess.run(
+        [str(RUNTIME), "--transform-only", str(cfg)],
+        env={"K3SAMD_SYSFS_ROOT": str(sysfs),
+             "K3SAMD_DEV_ROOT": str(tmp_path / "nodev"),
+             "K3SAMD_ROCM_ROOT": str(rocm)},
+        check=True, capture_output=True, timeout=60)
+    out = json.loads(cfg.read_text())
+    rocm_mounts = [m for m in out["mounts"]
+                   if m["destination"] == "/opt/rocm"]
+    assert len(rocm_mounts) == 1
+    assert rocm_mounts[0]["source"] == str(rocm)
+    assert "ro" in rocm_mounts[0]["options"]
+
+
+def test_cpu_only_node_injects_nothing(tmp_path):
+    out, _ = transform(tmp_path, BASE_SPEC, n_gpus=0)
+    assert "/dev/kfd" not in json.dumps(out)
+
+
+def test_malformed_config_fails(tmp_path):
+    cfg = tmp_path / "config.json"
+    cfg.write_text("{not json")
+    proc = subprocess.run([str(RUNTIME), "--transform-only", str(cfg)],
+                          capture_output=True, text=True, timeout=60)
+    assert proc.returncode != 0
