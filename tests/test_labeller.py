@@ -1,0 +1,63 @@
+"""Node labeller tests (replaces NFD/GFD roles: README.md:97-103,
+values.yaml:1-2; label surface mirrors nvidia.com/gpu.* incl. the
+nodeSelector example nvidia-smi.yaml:6-7)."""
+
+import json
+import subprocess
+from pathlib import Path
+
+import pytest
+
+from sysfs_builder import build_tree
+
+REPO = Path(__file__).resolve().parent.parent
+LABELLER = REPO / "native" / "bin" / "k3samd-node-labeller"
+
+
+@pytest.fixture(scope="session", autouse=True)
+def build_native():
+    subprocess.run(["make", "-C", str(REPO / "native"), "-j8"], check=True,
+                   capture_output=True)
+
+
+def run_labeller(sysfs_root, *args):
+    return subprocess.run(
+        [str(LABELLER), *args],
+        env={"K3SAMD_SYSFS_ROOT": str(sysfs_root)},
+        capture_output=True, text=True, timeout=60)
+
+
+def test_labels_eight_gpu(tmp_path):
+    root = build_tree(tmp_path / "sys", n_gpus=8)
+    out = run_labeller(root, "--json")
+    labels = json.loads(out.stdout)
+    assert labels["amd.com/gpu.present"] == "true"
+    assert labels["amd.com/gpu.count"] == "8"
+    assert labels["amd.com/gpu.arch"] == "gfx950"
+    assert labels["amd.com/gpu.family"] == "CDNA4"
+    assert labels["amd.com/gpu.product"] == "AMD-Instinct-MI355X"
+    assert labels["amd.com/gpu.vram"] == "294912Mi"  # 288 GiB
+    assert labels["amd.com/gpu.xgmi-links"] == "7"
+    assert labels["amd.com/gpu.cu-count"] == "256"
+    assert labels["amd.com/gpu.driver-version"] == "6.12.12"
+
+
+def test_features_file(tmp_path):
+    root = build_tree(tmp_path / "sys", n_gpus=2)
+    ff = tmp_path / "features.d" / "k3samd"
+    ff.parent.mkdir()
+    out = run_labeller(root, "--oneshot", "--features-file", str(ff))
+    assert out.returncode == 0
+    lines = dict(l.split("=", 1) for l in ff.read_text().splitlines())
+    assert lines["amd.com/gpu.present"] == "true"
+    assert lines["amd.com/gpu.count"] == "2"
+    assert not (tmp_path / "features.d" / "k3samd.tmp").exists()
+
+
+def test_cpu_only_no_labels(tmp_path):
+    root = build_tree(tmp_path / "sys", n_gpus=0)
+    ff = tmp_path / "k3samd"
+    run_labeller(root, "--oneshot", "--features-file", str(ff))
+    assert ff.read_text() == ""  # empty feature file => no gpu labels
+    out = run_labeller(root, "--json")
+    assert json.loads(out.stdout) == {}
