@@ -1,3 +1,4 @@
+#include "hip/hip_runtime.h"
 // k3samd_kernels.hip — torch bindings for the CDNA4 smoke/bench kernels.
 //
 // Device code lives in stream_kernels.h (shared with the standalone in-pod

@@ -1,0 +1,153 @@
+// CDNA4 (gfx950) device kernels shared by the torch extension
+// (k3samd_kernels.hip) and the standalone in-pod smoke binary
+// (native/hipsmoke/mi_stream.hip). Pure HIP — no torch dependency.
+//
+// Design (MI355X-first):
+//  * 16 B/lane (float4 ext-vector) loads/stores — 1 KiB per wave64 vector
+//    memory instruction, the HBM3E coalescing sweet spot.
+//  * non-temporal variants for the >L3 streaming regime (measured faster:
+//    6.40 TB/s vs 5.83 TB/s plain triad on MI355X silicon).
+//  * MFMA kernels use the gfx950 v_mfma_f32_16x16x32_bf16 shape; fragment
+//    layout validated on silicon (lane l: A[l&15][(l>>4)*8+j]).
+
+#pragma once
+
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+
+namespace k3samd_kern {
+
+constexpr int kThreadsPerBlock = 256;  // 4 waves of 64
+
+using f4 = __attribute__((ext_vector_type(4))) float;
+
+__device__ __forceinline__ f4 triad_op(const f4 b, const f4 c, float s) {
+  return b + s * c;
+}
+
+template <bool NT>
+__global__ void stream_triad_kernel(f4* __restrict__ a, const f4* __restrict__ b,
+                                    const f4* __restrict__ c, float s,
+                                    int64_t n4) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n4) return;
+  if constexpr (NT) {
+    f4 bv = __builtin_nontemporal_load(&b[i]);
+    f4 cv = __builtin_nontemporal_load(&c[i]);
+    __builtin_nontemporal_store(triad_op(bv, cv, s), &a[i]);
+  } else {
+    a[i] = triad_op(b[i], c[i], s);
+  }
+}
+
+template <bool NT>
+__global__ void stream_copy_kernel(f4* __restrict__ a, const f4* __restrict__ b,
+                                   int64_t n4) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n4) return;
+  if constexpr (NT) {
+    __builtin_nontemporal_store(__builtin_nontemporal_load(&b[i]), &a[i]);
+  } else {
+    a[i] = b[i];
+  }
+}
+
+template <bool NT>
+__global__ void stream_scale_kernel(f4* __restrict__ a, const f4* __restrict__ c,
+                                    float s, int64_t n4) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n4) return;
+  if constexpr (NT) {
+    __builtin_nontemporal_store(s * __builtin_nontemporal_load(&c[i]), &a[i]);
+  } else {
+    a[i] = s * c[i];
+  }
+}
+
+template <bool NT>
+__global__ void stream_add_kernel(f4* __restrict__ a, const f4* __restrict__ b,
+                                  const f4* __restrict__ c, int64_t n4) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n4) return;
+  if constexpr (NT) {
+    f4 bv = __builtin_nontemporal_load(&b[i]);
+    f4 cv = __builtin_nontemporal_load(&c[i]);
+    __builtin_nontemporal_store(bv + cv, &a[i]);
+  } else {
+    a[i] = b[i] + c[i];
+  }
+}
+
+// ---------------------------------------------------------------------------
+// MFMA kernels (device code only selected when compiling for gfx950).
+// ---------------------------------------------------------------------------
+
+#if defined(__gfx950__)
+#define K3_HAS_MFMA 1
+using bf16x8 = __attribute__((ext_vector_type(8))) short;   // 8 bf16 = 4 VGPR
+using f32x4 = __attribute__((ext_vector_type(4))) float;    // C/D for 16x16
+#else
+#define K3_HAS_MFMA 0
+#endif
+
+// MFMA throughput: 4 independent accumulators per wave (> the 2 needed to
+// reach the 32-cycle issue rate of v_mfma_f32_16x16x32_bf16, dependent
+// latency 40 < 2x32). FLOPs per MFMA = 2*16*16*32 = 16384.
+static __global__ void mfma_throughput_kernel(float* __restrict__ out,
+                                              int iters) {
+#if K3_HAS_MFMA
+  bf16x8 a, b;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    a[j] = (short)(0x3f80 + ((threadIdx.x + j) & 7));
+    b[j] = (short)(0x3f00 + ((threadIdx.x * 3 + j) & 7));
+  }
+  f32x4 acc0 = {0.f, 0.f, 0.f, 0.f}, acc1 = acc0, acc2 = acc0, acc3 = acc0;
+  for (int i = 0; i < iters; ++i) {
+    acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc0, 0, 0, 0);
+    acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc1, 0, 0, 0);
+    acc2 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc2, 0, 0, 0);
+    acc3 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc3, 0, 0, 0);
+  }
+  float r = acc0[0] + acc1[1] + acc2[2] + acc3[3];
+  if (threadIdx.x == 0) out[blockIdx.x] = r;  // keep the work alive
+#else
+  if (threadIdx.x == 0) out[blockIdx.x] = -1.f;
+#endif
+}
+
+// Single-tile D[16x16] = A[16x32] * B[32x16] through one
+// v_mfma_f32_16x16x32_bf16 for numerics validation.
+//   layout 0 (validated on MI355X): lane l holds A[l&15][(l>>4)*8 + j]
+//   layout 1 (rejected candidate) : A[l&15][(l>>4)*4 + (j&3) + 16*(j>>2)]
+// B mirrors A with row/col swapped; C/D: col=lane&15, row=(lane>>4)*4+reg.
+static __global__ void mfma_gemm16_kernel(const uint16_t* __restrict__ A,
+                                          const uint16_t* __restrict__ B,
+                                          float* __restrict__ D, int layout) {
+#if K3_HAS_MFMA
+  const int l = threadIdx.x;  // one wave
+  bf16x8 af, bf;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    int k = (layout == 0) ? ((l >> 4) * 8 + j)
+                          : ((l >> 4) * 4 + (j & 3) + 16 * (j >> 2));
+    af[j] = (short)A[(l & 15) * 32 + k];
+    bf[j] = (short)B[k * 16 + (l & 15)];
+  }
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf, acc, 0, 0, 0);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    D[((l >> 4) * 4 + r) * 16 + (l & 15)] = acc[r];
+  }
+#else
+  (void)A; (void)B; (void)D; (void)layout;
+#endif
+}
+
+inline int64_t stream_grid(int64_t n4) {
+  return (n4 + kThreadsPerBlock - 1) / kThreadsPerBlock;
+}
+
+}  // namespace k3samd_kern

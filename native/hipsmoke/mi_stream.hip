@@ -1,0 +1,187 @@
+// mi-stream — standalone CDNA4 STREAM + MFMA smoke payload for the
+// validation pod.
+//
+// The MI355X-native replacement for the reference's in-pod `nvidia-smi`
+// payload (/root/reference/nvidia-smi.yaml:13) with actual GPU work: an
+// HBM3E STREAM suite (copy/scale/add/triad, plain + non-temporal) and an
+// MFMA matrix-core warm-up that lights up all 8 XCDs, printing a
+// golden-output-style table (the analog of README.md:137-156) plus one
+// machine-readable JSON line. No torch dependency — links only the HIP
+// runtime, so the smoke container stays small.
+//
+// Build: hipcc --offload-arch=gfx950 -O3 mi_stream.hip -o mi-stream
+
+#include <cstdio>
+#include <cstring>
+#include <string>
+#include <vector>
+
+#include "../../k3samd/ops/hip/stream_kernels.h"
+
+#define HIP_CHECK(x)                                                         \
+  do {                                                                       \
+    hipError_t err_ = (x);                                                   \
+    if (err_ != hipSuccess) {                                                \
+      std::fprintf(stderr, "mi-stream: %s failed: %s\n", #x,                 \
+                   hipGetErrorString(err_));                                 \
+      return 1;                                                              \
+    }                                                                        \
+  } while (0)
+
+using k3samd_kern::f4;
+
+namespace {
+
+struct Timing {
+  float best_ms = 1e30f;
+};
+
+double gbps(double bytes, float ms) { return bytes / (ms * 1e6); }
+
+}  // namespace
+
+int main(int argc, char** argv) {
+  int64_t mib = 1024;
+  int iters = 20;
+  int device = 0;
+  bool do_mfma = true;
+  for (int i = 1; i < argc; ++i) {
+    if (!std::strcmp(argv[i], "--mib") && i + 1 < argc)
+      mib = std::atoll(argv[++i]);
+    else if (!std::strcmp(argv[i], "--iters") && i + 1 < argc)
+      iters = std::atoi(argv[++i]);
+    else if (!std::strcmp(argv[i], "--device") && i + 1 < argc)
+      device = std::atoi(argv[++i]);
+    else if (!std::strcmp(argv[i], "--no-mfma"))
+      do_mfma = false;
+    else {
+      std::printf("mi-stream [--mib N] [--iters N] [--device D] [--no-mfma]\n");
+      return !std::strcmp(argv[i], "--help") ? 0 : 2;
+    }
+  }
+
+  int ndev = 0;
+  HIP_CHECK(hipGetDeviceCount(&ndev));
+  if (ndev == 0) {
+    std::fprintf(stderr, "mi-stream: no GPUs visible\n");
+    return 1;
+  }
+  HIP_CHECK(hipSetDevice(device));
+  hipDeviceProp_t prop;
+  HIP_CHECK(hipGetDeviceProperties(&prop, device));
+
+  std::printf("mi-stream: %s (%s), %d visible GPU(s), %d CUs, %.0f GiB VRAM\n",
+              prop.name, prop.gcnArchName, ndev, prop.multiProcessorCount,
+              (double)prop.totalGlobalMem / (1 << 30));
+
+  const int64_t n = mib * (1 << 20) / 4;  // fp32 elements
+  const int64_t n4 = n / 4;
+  const double buf_bytes = (double)n * 4;
+  f4 *a, *b, *c;
+  HIP_CHECK(hipMalloc(&a, buf_bytes));
+  HIP_CHECK(hipMalloc(&b, buf_bytes));
+  HIP_CHECK(hipMalloc(&c, buf_bytes));
+  HIP_CHECK(hipMemset(b, 0x3c, buf_bytes));
+  HIP_CHECK(hipMemset(c, 0x3d, buf_bytes));
+
+  hipEvent_t ev0, ev1;
+  HIP_CHECK(hipEventCreate(&ev0));
+  HIP_CHECK(hipEventCreate(&ev1));
+  dim3 grid((uint32_t)k3samd_kern::stream_grid(n4));
+  dim3 block(k3samd_kern::kThreadsPerBlock);
+  const float s = 2.5f;
+
+  struct Row {
+    const char* name;
+    double bytes;
+    Timing plain, nt;
+  };
+  Row rows[] = {
+      {"copy", 2 * buf_bytes, {}, {}},
+      {"scale", 2 * buf_bytes, {}, {}},
+      {"add", 3 * buf_bytes, {}, {}},
+      {"triad", 3 * buf_bytes, {}, {}},
+  };
+
+  auto launch = [&](int op, bool nt) {
+    using namespace k3samd_kern;
+    if (nt) {
+      switch (op) {
+        case 0: hipLaunchKernelGGL(stream_copy_kernel<true>, grid, block, 0, 0, a, b, n4); break;
+        case 1: hipLaunchKernelGGL(stream_scale_kernel<true>, grid, block, 0, 0, a, c, s, n4); break;
+        case 2: hipLaunchKernelGGL(stream_add_kernel<true>, grid, block, 0, 0, a, b, c, n4); break;
+        case 3: hipLaunchKernelGGL(stream_triad_kernel<true>, grid, block, 0, 0, a, b, c, s, n4); break;
+      }
+    } else {
+      switch (op) {
+        case 0: hipLaunchKernelGGL(stream_copy_kernel<false>, grid, block, 0, 0, a, b, n4); break;
+        case 1: hipLaunchKernelGGL(stream_scale_kernel<false>, grid, block, 0, 0, a, c, s, n4); break;
+        case 2: hipLaunchKernelGGL(stream_add_kernel<false>, grid, block, 0, 0, a, b, c, n4); break;
+        case 3: hipLaunchKernelGGL(stream_triad_kernel<false>, grid, block, 0, 0, a, b, c, s, n4); break;
+      }
+    }
+  };
+
+  for (int op = 0; op < 4; ++op) {
+    for (int nt = 0; nt < 2; ++nt) {
+      launch(op, nt);  // warm
+      launch(op, nt);
+      HIP_CHECK(hipDeviceSynchronize());
+      Timing& t = nt ? rows[op].nt : rows[op].plain;
+      for (int it = 0; it < iters; ++it) {
+        HIP_CHECK(hipEventRecord(ev0));
+        launch(op, nt);
+        HIP_CHECK(hipEventRecord(ev1));
+        HIP_CHECK(hipEventSynchronize(ev1));
+        float ms;
+        HIP_CHECK(hipEventElapsedTime(&ms, ev0, ev1));
+        if (ms < t.best_ms) t.best_ms = ms;
+      }
+    }
+  }
+
+  std::printf("+--------+-------------------+-------------------+\n");
+  std::printf("| STREAM | plain GB/s        | non-temporal GB/s |\n");
+  std::printf("+--------+-------------------+-------------------+\n");
+  for (auto& r : rows) {
+    std::printf("| %-6s | %17.1f | %17.1f |\n", r.name,
+                gbps(r.bytes, r.plain.best_ms), gbps(r.bytes, r.nt.best_ms));
+  }
+  std::printf("+--------+-------------------+-------------------+\n");
+
+  double mfma_tf = 0;
+  if (do_mfma) {
+    const int blocks = 2048, mf_iters = 4096;
+    float* out;
+    HIP_CHECK(hipMalloc(&out, blocks * sizeof(float)));
+    hipLaunchKernelGGL(k3samd_kern::mfma_throughput_kernel, dim3(blocks),
+                       block, 0, 0, out, 256);  // warm
+    HIP_CHECK(hipDeviceSynchronize());
+    HIP_CHECK(hipEventRecord(ev0));
+    hipLaunchKernelGGL(k3samd_kern::mfma_throughput_kernel, dim3(blocks),
+                       block, 0, 0, out, mf_iters);
+    HIP_CHECK(hipEventRecord(ev1));
+    HIP_CHECK(hipEventSynchronize(ev1));
+    float ms;
+    HIP_CHECK(hipEventElapsedTime(&ms, ev0, ev1));
+    double flops = (double)blocks * 4 * 4 * 16384.0 * mf_iters;
+    mfma_tf = flops / (ms * 1e9);
+    std::printf("| MFMA bf16 (v_mfma_f32_16x16x32_bf16): %8.1f TFLOP/s    |\n",
+                mfma_tf);
+    std::printf("+--------+-------------------+-------------------+\n");
+    HIP_CHECK(hipFree(out));
+  }
+
+  double triad_best =
+      gbps(rows[3].bytes, std::min(rows[3].plain.best_ms, rows[3].nt.best_ms));
+  std::printf(
+      "{\"payload\": \"mi-stream\", \"gpu\": \"%s\", \"arch\": \"%s\", "
+      "\"buffer_MiB\": %lld, \"triad_gbps\": %.1f, \"mfma_bf16_tflops\": "
+      "%.1f}\n",
+      prop.name, prop.gcnArchName, (long long)mib, triad_best, mfma_tf);
+
+  HIP_CHECK(hipFree(a));
+  HIP_CHECK(hipFree(b));
+  HIP_CHECK(hipFree(c));
+  return 0;
+}
