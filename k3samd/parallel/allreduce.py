@@ -1,0 +1,100 @@
+"""RCCL-over-xGMI collective smoke path (torch.distributed).
+
+The multi-process counterpart of native/hipsmoke/mi_allreduce.hip: one rank
+per GPU over the "nccl" backend (RCCL on ROCm), or "gloo" on CPU where the
+same code doubles as the distributed-path unit test.
+
+Run inside a pod requesting `amd.com/gpu: N`:
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 -m k3samd.parallel.allreduce
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+
+def busbw_sweep(sizes_bytes, iters=20, device=None, group=None):
+    """All-reduce each size `iters` times; returns [{bytes, algbw, busbw}].
+
+    busbw = 2*(n-1)/n * algbw — the standard ring-equivalent bus bandwidth
+    normalization, comparable across world sizes.
+    """
+    import torch
+    import torch.distributed as dist
+
+    world = dist.get_world_size(group)
+    rows = []
+    for bytes_ in sizes_bytes:
+        n = max(int(bytes_) // 4, 1)
+        t = torch.rand(n, dtype=torch.float32, device=device)
+        for _ in range(3):
+            dist.all_reduce(t, group=group)
+        if device is not None and str(device).startswith("cuda"):
+            torch.cuda.synchronize(device)
+        dist.barrier(group)
+        t0 = time.perf_counter()
+        for _ in range(iters):
+            dist.all_reduce(t, group=group)
+        if device is not None and str(device).startswith("cuda"):
+            torch.cuda.synchronize(device)
+        elapsed = time.perf_counter() - t0
+        # max over ranks so the report reflects the slowest participant
+        e = torch.tensor([elapsed], dtype=torch.float64)
+        dist.all_reduce(e, op=dist.ReduceOp.MAX, group=group)
+        per_op = float(e[0]) / iters
+        algbw = n * 4 / per_op / 1e9
+        busbw = algbw * 2 * (world - 1) / world
+        rows.append({"bytes": n * 4, "algbw_gbps": round(algbw, 2),
+                     "busbw_gbps": round(busbw, 2)})
+    return rows
+
+
+def main(argv=None) -> int:
+    import torch
+    import torch.distributed as dist
+
+    p = argparse.ArgumentParser(description=__doc__)
+    p.add_argument("--min-mib", type=int, default=1)
+    p.add_argument("--max-mib", type=int, default=1024)
+    p.add_argument("--iters", type=int, default=20)
+    p.add_argument("--backend", default="auto")
+    args = p.parse_args(argv)
+
+    backend = args.backend
+    if backend == "auto":
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29572")
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    dist.init_process_group(backend=backend, rank=rank, world_size=world)
+
+    device = None
+    if backend == "nccl":
+        torch.cuda.set_device(local_rank)
+        device = torch.device("cuda", local_rank)
+
+    sizes = []
+    b = args.min_mib << 20
+    while b <= args.max_mib << 20:
+        sizes.append(b)
+        b *= 4
+    rows = busbw_sweep(sizes, iters=args.iters, device=device)
+
+    if rank == 0:
+        for r in rows:
+            print(f"{r['bytes']:>12d} B  algbw {r['algbw_gbps']:8.2f} GB/s"
+                  f"  busbw {r['busbw_gbps']:8.2f} GB/s")
+        print(json.dumps({"payload": "allreduce", "backend": backend,
+                          "world_size": world, "rows": rows}))
+    dist.destroy_process_group()
+    return 0
+
+
+if __name__ == "__main__":
+    raise SystemExit(main())
