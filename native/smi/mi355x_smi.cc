@@ -13,6 +13,7 @@
 
 #include <cstdio>
 #include <cstring>
+#include <dirent.h>
 #include <fstream>
 #include <sstream>
 #include <string>
@@ -54,14 +55,18 @@ CardStats card_stats(const std::string& sysfs_root, int card_index) {
   st.vram_total = read_u64(dev + "/mem_info_vram_total");
   std::string busy = read_trim(dev + "/gpu_busy_percent");
   if (!busy.empty()) st.busy_percent = std::strtol(busy.c_str(), nullptr, 10);
-  // first hwmon temp input if present
-  for (int h = 0; h < 8; ++h) {
-    std::string t = read_trim(dev + "/hwmon/hwmon" + std::to_string(h) +
-                              "/temp1_input");
-    if (!t.empty()) {
-      st.temp_mc = std::strtol(t.c_str(), nullptr, 10);
-      break;
+  // scan hwmon/* (the index is global, not per-card, on real systems)
+  std::string hwdir = dev + "/hwmon";
+  if (DIR* d = ::opendir(hwdir.c_str())) {
+    while (struct dirent* e = ::readdir(d)) {
+      if (e->d_name[0] == '.') continue;
+      std::string t = read_trim(hwdir + "/" + e->d_name + "/temp1_input");
+      if (!t.empty()) {
+        st.temp_mc = std::strtol(t.c_str(), nullptr, 10);
+        break;
+      }
     }
+    ::closedir(d);
   }
   return st;
 }

@@ -101,7 +101,15 @@ std::string default_sysfs_root() {
 
 Topology enumerate_topology(const std::string& sysfs_root) {
   Topology topo;
+  // DKMS builds publish /sys/module/amdgpu/version; the in-tree driver does
+  // not — fall back to the running kernel release (the driver version IS
+  // the kernel version for in-tree amdgpu).
   topo.driver_version = trim(read_file(sysfs_root + "/module/amdgpu/version"));
+  if (topo.driver_version.empty() &&
+      is_dir(sysfs_root + "/module/amdgpu")) {
+    std::string rel = trim(read_file("/proc/sys/kernel/osrelease"));
+    if (!rel.empty()) topo.driver_version = "in-tree/" + rel;
+  }
 
   const std::string nodes_dir = sysfs_root + "/class/kfd/kfd/topology/nodes";
   if (!is_dir(nodes_dir)) return topo;  // no KFD => no GPUs (CPU-only node)
@@ -170,8 +178,19 @@ Topology enumerate_topology(const std::string& sysfs_root) {
         std::string ue = read_file(link + "/uevent");
         if (ue.find(g.pci_bdf) != std::string::npos) {
           g.card_index = idx;
+          // The KFD `name` file on real MI355X silicon reports the IP
+          // discovery string ("ip discovery"), not a marketing name; prefer
+          // the PCI product_name the amdgpu driver publishes when present.
+          std::string product = trim(read_file(link + "/product_name"));
+          if (!product.empty() &&
+              (g.name.empty() || g.name == "ip discovery")) {
+            g.name = product;
+          }
           break;
         }
+      }
+      if (g.name.empty() || g.name == "ip discovery") {
+        g.name = "AMD GPU " + g.gfx_arch();
       }
     }
 
