@@ -4,6 +4,8 @@
 #include <chrono>
 #include <cstdio>
 #include <set>
+#include <sys/stat.h>
+#include <thread>
 
 #include "../common/miniyaml.h"
 
@@ -173,6 +175,9 @@ GrpcStatus DevicePlugin::handle_allocate(const std::string& req,
     cr.envs["K3SAMD_VISIBLE_DEVICES"] = visible;
     cr.envs["K3SAMD_RENDER_MINORS"] = minors;
     cr.annotations["k3samd.ai/allocated-gpus"] = visible;
+    std::fprintf(stderr,
+                 "deviceplugin: Allocate %zu id(s) -> gpus [%s] minors [%s]\n",
+                 ids.size(), visible.c_str(), minors.c_str());
     crs.push_back(std::move(cr));
   }
   resp = dp::encode_allocate_response(crs);
@@ -291,24 +296,35 @@ bool DevicePlugin::serve(const std::string& plugin_sock,
     return false;
   }
 
+  plugin_sock_ = plugin_sock;
+  kubelet_sock_ = kubelet_sock;
+  stopping_.store(false);
+
   if (!kubelet_sock.empty()) {
-    dp::RegisterRequest rr;
-    size_t slash = plugin_sock.find_last_of('/');
-    rr.endpoint =
-        slash == std::string::npos ? plugin_sock : plugin_sock.substr(slash + 1);
-    rr.resource_name = advertised_resource();
-    auto res = grpc_unary_call(kubelet_sock, dp::kRegisterPath, rr.encode());
-    if (!res.transport_ok || res.grpc_status != 0) {
-      std::fprintf(stderr,
-                   "deviceplugin: Register with kubelet failed: %s (grpc=%d %s)\n",
-                   res.error.c_str(), res.grpc_status,
-                   res.grpc_message.c_str());
+    if (!register_with_kubelet()) {
       server_.stop();
       return false;
     }
+    // kubelet drops all plugin registrations when it restarts: watch the
+    // registration socket's inode and re-register when it changes (the
+    // same recovery loop production device plugins implement).
+    reregister_thread_ = std::thread([this] {
+      struct stat st{};
+      ino_t last_ino = ::stat(kubelet_sock_.c_str(), &st) == 0 ? st.st_ino : 0;
+      while (!stopping_.load()) {
+        std::this_thread::sleep_for(std::chrono::milliseconds(500));
+        if (stopping_.load()) break;
+        struct stat now{};
+        if (::stat(kubelet_sock_.c_str(), &now) != 0) continue;
+        if (now.st_ino != last_ino) {
+          std::fprintf(stderr,
+                       "deviceplugin: kubelet socket changed, re-registering\n");
+          if (register_with_kubelet()) last_ino = now.st_ino;
+        }
+      }
+    });
   }
 
-  stopping_.store(false);
   if (health_poll_ms > 0) {
     health_thread_ = std::thread([this, health_poll_ms] {
       while (!stopping_.load()) {
@@ -321,10 +337,27 @@ bool DevicePlugin::serve(const std::string& plugin_sock,
   return true;
 }
 
+bool DevicePlugin::register_with_kubelet() {
+  dp::RegisterRequest rr;
+  size_t slash = plugin_sock_.find_last_of('/');
+  rr.endpoint = slash == std::string::npos ? plugin_sock_
+                                           : plugin_sock_.substr(slash + 1);
+  rr.resource_name = advertised_resource();
+  auto res = grpc_unary_call(kubelet_sock_, dp::kRegisterPath, rr.encode());
+  if (!res.transport_ok || res.grpc_status != 0) {
+    std::fprintf(stderr,
+                 "deviceplugin: Register with kubelet failed: %s (grpc=%d %s)\n",
+                 res.error.c_str(), res.grpc_status, res.grpc_message.c_str());
+    return false;
+  }
+  return true;
+}
+
 void DevicePlugin::stop() {
   stopping_.store(true);
   gen_cv_.notify_all();
   if (health_thread_.joinable()) health_thread_.join();
+  if (reregister_thread_.joinable()) reregister_thread_.join();
   server_.stop();
 }
 

@@ -86,8 +86,14 @@ class DevicePlugin {
   uint64_t generation_ = 0;  // bumped on device-list changes
   std::condition_variable gen_cv_;
 
+  // register with kubelet; returns false on failure
+  bool register_with_kubelet();
+
   GrpcServer server_;
   std::thread health_thread_;
+  std::thread reregister_thread_;
+  std::string plugin_sock_;
+  std::string kubelet_sock_;
   std::atomic<bool> stopping_{false};
 };
 

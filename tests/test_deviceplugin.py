@@ -38,6 +38,8 @@ class FakeKubelet:
     """Records Register calls, like kubelet's Registration service."""
 
     def __init__(self, sock_path):
+        # like the real kubelet, claim the socket path on startup
+        Path(sock_path).unlink(missing_ok=True)
         self.requests = []
         self.event = threading.Event()
         handler = grpc.method_handlers_generic_handler(
@@ -270,6 +272,20 @@ def test_health_transition_pushes_update(harness):
     second = pb.decode_list_and_watch(next(stream))
     by_health = sorted(d["health"] for d in second)
     assert by_health == ["Healthy", "Unhealthy"]
+
+
+def test_reregister_after_kubelet_restart(harness):
+    """kubelet restarts drop plugin registrations; the plugin must notice
+    the new registration socket and Register again."""
+    h = harness(n_gpus=1, replicas=1)
+    assert h.kubelet.event.wait(5)
+    assert len(h.kubelet.requests) == 1
+    # simulate kubelet restart: new server process on the same socket path
+    h.kubelet.stop()
+    time.sleep(0.2)
+    h.kubelet = FakeKubelet(h.kubelet_sock)
+    assert h.kubelet.event.wait(10), "plugin did not re-register"
+    assert h.kubelet.requests[0]["resource_name"] == "amd.com/gpu"
 
 
 def test_cpu_only_zero_allocatable(tmp_path):
