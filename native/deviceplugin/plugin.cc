@@ -84,7 +84,6 @@ void DevicePlugin::refresh_topology() {
     }
   }
   ++generation_;
-  gen_cv_.notify_all();
 }
 
 std::vector<VirtualDevice> DevicePlugin::devices() const {
@@ -235,7 +234,6 @@ bool DevicePlugin::poll_health_once() {
   }
   if (changed) {
     ++generation_;
-    gen_cv_.notify_all();
   }
   return changed;
 }
@@ -249,16 +247,18 @@ GrpcStatus DevicePlugin::handle_list_and_watch(
   }
   if (!write(dp::encode_list_and_watch(current_device_list())))
     return GrpcStatus::Ok();
-  // push an update whenever the device list changes (kubelet keeps this
-  // stream open for the plugin's lifetime)
+  // Push an update whenever the device list changes (kubelet keeps this
+  // stream open for the plugin's lifetime). A bounded 100 ms poll keeps the
+  // concurrency trivial — health transitions are second-granularity events.
   while (!stopping_.load()) {
-    std::unique_lock<std::mutex> lk(mu_);
-    gen_cv_.wait_for(lk, std::chrono::seconds(1),
-                     [&] { return generation_ != seen_gen || stopping_.load(); });
-    if (stopping_.load()) break;
-    if (generation_ == seen_gen) continue;
-    seen_gen = generation_;
-    lk.unlock();
+    std::this_thread::sleep_for(std::chrono::milliseconds(100));
+    uint64_t gen;
+    {
+      std::lock_guard<std::mutex> lk(mu_);
+      gen = generation_;
+    }
+    if (gen == seen_gen) continue;
+    seen_gen = gen;
     if (!write(dp::encode_list_and_watch(current_device_list()))) break;
   }
   return GrpcStatus::Ok();
@@ -355,7 +355,6 @@ bool DevicePlugin::register_with_kubelet() {
 
 void DevicePlugin::stop() {
   stopping_.store(true);
-  gen_cv_.notify_all();
   if (health_thread_.joinable()) health_thread_.join();
   if (reregister_thread_.joinable()) reregister_thread_.join();
   server_.stop();

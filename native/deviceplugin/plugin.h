@@ -9,7 +9,6 @@
 #pragma once
 
 #include <atomic>
-#include <condition_variable>
 #include <mutex>
 #include <string>
 #include <thread>
@@ -84,7 +83,6 @@ class DevicePlugin {
   Topology topo_;
   std::vector<VirtualDevice> devices_;
   uint64_t generation_ = 0;  // bumped on device-list changes
-  std::condition_variable gen_cv_;
 
   // register with kubelet; returns false on failure
   bool register_with_kubelet();

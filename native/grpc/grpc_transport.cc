@@ -221,22 +221,21 @@ void GrpcServer::add_server_stream(const std::string& path, StreamHandler h) {
 
 bool GrpcServer::start(const std::string& unix_path) {
   ::unlink(unix_path.c_str());
-  listen_fd_ = ::socket(AF_UNIX, SOCK_STREAM | SOCK_CLOEXEC, 0);
-  if (listen_fd_ < 0) return false;
+  int fd = ::socket(AF_UNIX, SOCK_STREAM | SOCK_CLOEXEC, 0);
+  if (fd < 0) return false;
   sockaddr_un addr{};
   addr.sun_family = AF_UNIX;
   if (unix_path.size() >= sizeof(addr.sun_path)) return false;
   std::strcpy(addr.sun_path, unix_path.c_str());
-  if (::bind(listen_fd_, (sockaddr*)&addr, sizeof(addr)) != 0) {
-    ::close(listen_fd_);
-    listen_fd_ = -1;
+  if (::bind(fd, (sockaddr*)&addr, sizeof(addr)) != 0) {
+    ::close(fd);
     return false;
   }
-  if (::listen(listen_fd_, 8) != 0) {
-    ::close(listen_fd_);
-    listen_fd_ = -1;
+  if (::listen(fd, 8) != 0) {
+    ::close(fd);
     return false;
   }
+  listen_fd_.store(fd);
   running_.store(true);
   stopping_.store(false);
   accept_thread_ = std::thread([this] { serve_loop(); });
@@ -244,21 +243,37 @@ bool GrpcServer::start(const std::string& unix_path) {
 }
 
 void GrpcServer::stop() {
-  if (!running_.load() && listen_fd_ < 0) return;
+  int fd = listen_fd_.exchange(-1);
+  if (!running_.load() && fd < 0) return;
   stopping_.store(true);
-  if (listen_fd_ >= 0) {
-    ::shutdown(listen_fd_, SHUT_RDWR);
-    ::close(listen_fd_);
-    listen_fd_ = -1;
+  if (fd >= 0) {
+    ::shutdown(fd, SHUT_RDWR);
+    ::close(fd);
   }
   {
     std::lock_guard<std::mutex> lk(conns_mu_);
     for (auto& c : conns_) c->close_fd();
   }
   if (accept_thread_.joinable()) accept_thread_.join();
-  // connection handler threads are detached but hold shared_ptrs; they exit
-  // promptly once the fd is shut down.
+  // join every connection/stream thread (their fds are shut down above, so
+  // they exit promptly; new stream threads can only be added by live
+  // connection threads, which this loop drains)
+  for (;;) {
+    std::thread t;
+    {
+      std::lock_guard<std::mutex> lk(threads_mu_);
+      if (threads_.empty()) break;
+      t = std::move(threads_.back());
+      threads_.pop_back();
+    }
+    if (t.joinable()) t.join();
+  }
   running_.store(false);
+}
+
+void GrpcServer::track_thread(std::thread t) {
+  std::lock_guard<std::mutex> lk(threads_mu_);
+  threads_.push_back(std::move(t));
 }
 
 namespace {
@@ -297,7 +312,9 @@ std::string find_header(const std::vector<Header>& hs, const std::string& k) {
 
 void GrpcServer::serve_loop() {
   while (!stopping_.load()) {
-    int cfd = ::accept(listen_fd_, nullptr, nullptr);
+    int lfd = listen_fd_.load();
+    if (lfd < 0) break;
+    int cfd = ::accept(lfd, nullptr, nullptr);
     if (cfd < 0) break;
     auto conn = std::make_shared<H2Conn>();
     conn->fd = cfd;
@@ -305,7 +322,7 @@ void GrpcServer::serve_loop() {
       std::lock_guard<std::mutex> lk(conns_mu_);
       conns_.insert(conn);
     }
-    std::thread([this, conn] {
+    track_thread(std::thread([this, conn] {
       // --- connection handshake ---
       char preface[kPrefaceLen];
       if (!read_full(conn->fd, preface, kPrefaceLen, 10000) ||
@@ -485,7 +502,7 @@ void GrpcServer::serve_loop() {
             auto sit = stream_.find(path);
             if (sit != stream_.end()) {
               StreamHandler handler = sit->second;
-              std::thread([conn, st, handler, req] {
+              track_thread(std::thread([conn, st, handler, req] {
                 send_response_headers(conn, st);
                 auto write = [conn, st](const std::string& msg) -> bool {
                   if (conn->closed.load() || st->cancelled.load()) return false;
@@ -494,7 +511,7 @@ void GrpcServer::serve_loop() {
                 GrpcStatus gs = handler(req, write);
                 if (!conn->closed.load() && !st->cancelled.load())
                   send_trailers(conn, st, gs, false);
-              }).detach();
+              }));
               continue;
             }
             send_trailers(conn, st, {12, "unknown method " + path}, true);
@@ -507,7 +524,7 @@ void GrpcServer::serve_loop() {
         std::lock_guard<std::mutex> lk(conns_mu_);
         conns_.erase(conn);
       }
-    }).detach();
+    }));
   }
 }
 

@@ -20,6 +20,7 @@
 #include <set>
 #include <string>
 #include <thread>
+#include <vector>
 
 namespace k3samd {
 
@@ -58,12 +59,17 @@ class GrpcServer {
 
   std::map<std::string, UnaryHandler> unary_;
   std::map<std::string, StreamHandler> stream_;
-  int listen_fd_ = -1;
+  std::atomic<int> listen_fd_{-1};
   std::thread accept_thread_;
   std::atomic<bool> running_{false};
   std::atomic<bool> stopping_{false};
   std::mutex conns_mu_;
   std::set<std::shared_ptr<struct H2Conn>> conns_;
+  // every spawned thread (connection readers + stream handlers) is tracked
+  // and joined in stop() so no thread outlives the server object
+  void track_thread(std::thread t);
+  std::mutex threads_mu_;
+  std::vector<std::thread> threads_;
 };
 
 struct UnaryCallResult {
