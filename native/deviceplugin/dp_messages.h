@@ -150,6 +150,7 @@ struct ContainerAllocateResponse {
   std::vector<Mount> mounts;                       // 2
   std::vector<DeviceSpec> devices;                 // 3
   std::map<std::string, std::string> annotations;  // 4
+  std::vector<std::string> cdi_devices;            // 5: CDIDevice{name=1}
 
   std::string encode() const {
     std::string out;
@@ -157,6 +158,11 @@ struct ContainerAllocateResponse {
     for (auto& m : mounts) pb::put_bytes(out, 2, m.encode());
     for (auto& d : devices) pb::put_bytes(out, 3, d.encode());
     for (auto& [k, v] : annotations) pb::put_map_entry(out, 4, k, v);
+    for (auto& n : cdi_devices) {
+      std::string cd;
+      pb::put_string(cd, 1, n);
+      pb::put_bytes(out, 5, cd);
+    }
     return out;
   }
 };

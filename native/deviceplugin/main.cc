@@ -41,6 +41,7 @@ int main(int argc, char** argv) {
   std::string kubelet_sock = "/var/lib/kubelet/device-plugins/kubelet.sock";
   int health_poll_ms = 5000;
   bool oneshot = false;
+  bool use_cdi = false;
 
   for (int i = 1; i < argc; ++i) {
     auto arg = [&](const char* name) -> const char* {
@@ -52,6 +53,7 @@ int main(int argc, char** argv) {
     else if (const char* v = arg("--kubelet-sock")) kubelet_sock = v;
     else if (const char* v = arg("--health-poll-ms")) health_poll_ms = std::atoi(v);
     else if (!std::strcmp(argv[i], "--no-register")) kubelet_sock.clear();
+    else if (!std::strcmp(argv[i], "--use-cdi")) use_cdi = true;
     else if (!std::strcmp(argv[i], "--oneshot")) oneshot = true;
     else if (!std::strcmp(argv[i], "--help") || !std::strcmp(argv[i], "-h")) {
       std::printf("k3samd-device-plugin [--config F] [--plugin-sock S] "
@@ -74,6 +76,7 @@ int main(int argc, char** argv) {
     }
   }
 
+  cfg.use_cdi = use_cdi;
   k3samd::DevicePlugin plugin(cfg, k3samd::default_sysfs_root());
 
   if (oneshot) {

@@ -154,17 +154,24 @@ GrpcStatus DevicePlugin::handle_allocate(const std::string& req,
       if (gi < 0) return {3, "unknown device id " + id};
       gpus.insert(gi);
     }
-    // /dev/kfd is the compute entry point, shared by all GPUs
-    cr.devices.push_back({"/dev/kfd", dev_root_ + "/kfd", "rw"});
     std::string visible, minors;
+    if (!cfg_.use_cdi) {
+      // /dev/kfd is the compute entry point, shared by all GPUs
+      cr.devices.push_back({"/dev/kfd", dev_root_ + "/kfd", "rw"});
+    }
     for (int gi : gpus) {
       const auto& g = topo_.gpus[gi];
-      std::string rnode =
-          "/dri/renderD" + std::to_string(g.drm_render_minor);
-      cr.devices.push_back({"/dev" + rnode, dev_root_ + rnode, "rw"});
-      if (g.card_index >= 0) {
-        std::string cnode = "/dri/card" + std::to_string(g.card_index);
-        cr.devices.push_back({"/dev" + cnode, dev_root_ + cnode, "rw"});
+      if (cfg_.use_cdi) {
+        // containerd resolves these against the k3samd-cdi-gen spec
+        cr.cdi_devices.push_back(cfg_.cdi_kind + "=" + g.stable_id());
+      } else {
+        std::string rnode =
+            "/dri/renderD" + std::to_string(g.drm_render_minor);
+        cr.devices.push_back({"/dev" + rnode, dev_root_ + rnode, "rw"});
+        if (g.card_index >= 0) {
+          std::string cnode = "/dri/card" + std::to_string(g.card_index);
+          cr.devices.push_back({"/dev" + cnode, dev_root_ + cnode, "rw"});
+        }
       }
       if (!visible.empty()) visible += ",";
       visible += g.stable_id();

@@ -28,6 +28,8 @@ struct PluginConfig {
   bool fail_requests_greater_than_one = false;
   int replicas = 1;                    // for resource `resource_name`
   std::string resource_name = "amd.com/gpu";
+  bool use_cdi = false;                // Allocate returns CDI device names
+  std::string cdi_kind = "amd.com/gpu";
 
   // Parse the `version: v1 / flags / sharing.timeSlicing` YAML document.
   static bool from_yaml(const std::string& text, PluginConfig& out,
