@@ -6,6 +6,7 @@
 #include <set>
 #include <sys/stat.h>
 #include <thread>
+#include <tuple>
 
 #include "../common/miniyaml.h"
 
@@ -316,17 +317,25 @@ bool DevicePlugin::serve(const std::string& plugin_sock,
     // registration socket's inode and re-register when it changes (the
     // same recovery loop production device plugins implement).
     reregister_thread_ = std::thread([this] {
+      // identity = (inode, ctime): tmpfs reuses inode numbers on an
+      // unlink+rebind, so the inode alone can miss a kubelet restart
+      auto ident = [](const struct stat& st) {
+        return std::make_tuple(st.st_ino, st.st_ctim.tv_sec,
+                               st.st_ctim.tv_nsec);
+      };
       struct stat st{};
-      ino_t last_ino = ::stat(kubelet_sock_.c_str(), &st) == 0 ? st.st_ino : 0;
+      auto last = ::stat(kubelet_sock_.c_str(), &st) == 0
+                      ? ident(st)
+                      : std::make_tuple((ino_t)0, (time_t)0, 0l);
       while (!stopping_.load()) {
         std::this_thread::sleep_for(std::chrono::milliseconds(500));
         if (stopping_.load()) break;
         struct stat now{};
         if (::stat(kubelet_sock_.c_str(), &now) != 0) continue;
-        if (now.st_ino != last_ino) {
+        if (ident(now) != last) {
           std::fprintf(stderr,
                        "deviceplugin: kubelet socket changed, re-registering\n");
-          if (register_with_kubelet()) last_ino = now.st_ino;
+          if (register_with_kubelet()) last = ident(now);
         }
       }
     });
