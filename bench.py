@@ -125,7 +125,8 @@ def main(argv=None) -> int:
                 run_step(nt)
             torch.cuda.synchronize()
             timings.append(time.perf_counter() - t0)
-        t = torch.tensor(timings, dtype=torch.float64)
+        # nccl collectives need device tensors
+        t = torch.tensor(timings, dtype=torch.float64, device=device)
         if dist is not None:
             dist.all_reduce(t)  # sum over ranks -> same decision everywhere
         use_nt = bool(t[1] < t[0])
@@ -148,7 +149,7 @@ def main(argv=None) -> int:
     elapsed = time.perf_counter() - t0
     barrier()
 
-    t = torch.tensor([elapsed], dtype=torch.float64)
+    t = torch.tensor([elapsed], dtype=torch.float64, device=device)
     if dist is not None:
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
     t_max = float(t[0])

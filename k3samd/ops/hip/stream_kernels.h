@@ -79,6 +79,41 @@ __global__ void stream_add_kernel(f4* __restrict__ a, const f4* __restrict__ b,
   }
 }
 
+// Grid-stride triad with U float4-elements per thread per pass — used by
+// the tuning sweep (mi-stream --tune) to probe block-size/unroll space.
+template <bool NT, int U>
+__global__ void stream_triad_gs_kernel(f4* __restrict__ a,
+                                       const f4* __restrict__ b,
+                                       const f4* __restrict__ c, float s,
+                                       int64_t n4) {
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  for (; i + (U - 1) * stride < n4; i += U * stride) {
+    f4 bv[U], cv[U];
+#pragma unroll
+    for (int u = 0; u < U; ++u) {
+      if constexpr (NT) {
+        bv[u] = __builtin_nontemporal_load(&b[i + u * stride]);
+        cv[u] = __builtin_nontemporal_load(&c[i + u * stride]);
+      } else {
+        bv[u] = b[i + u * stride];
+        cv[u] = c[i + u * stride];
+      }
+    }
+#pragma unroll
+    for (int u = 0; u < U; ++u) {
+      if constexpr (NT) {
+        __builtin_nontemporal_store(triad_op(bv[u], cv[u], s),
+                                    &a[i + u * stride]);
+      } else {
+        a[i + u * stride] = triad_op(bv[u], cv[u], s);
+      }
+    }
+  }
+  // tail
+  for (; i < n4; i += stride) a[i] = triad_op(b[i], c[i], s);
+}
+
 // ---------------------------------------------------------------------------
 // MFMA kernels (device code only selected when compiling for gfx950).
 // ---------------------------------------------------------------------------

@@ -45,6 +45,7 @@ int main(int argc, char** argv) {
   int iters = 20;
   int device = 0;
   bool do_mfma = true;
+  bool tune = false;
   for (int i = 1; i < argc; ++i) {
     if (!std::strcmp(argv[i], "--mib") && i + 1 < argc)
       mib = std::atoll(argv[++i]);
@@ -54,8 +55,11 @@ int main(int argc, char** argv) {
       device = std::atoi(argv[++i]);
     else if (!std::strcmp(argv[i], "--no-mfma"))
       do_mfma = false;
+    else if (!std::strcmp(argv[i], "--tune"))
+      tune = true;
     else {
-      std::printf("mi-stream [--mib N] [--iters N] [--device D] [--no-mfma]\n");
+      std::printf("mi-stream [--mib N] [--iters N] [--device D] [--no-mfma]"
+                  " [--tune]\n");
       return !std::strcmp(argv[i], "--help") ? 0 : 2;
     }
   }
@@ -70,8 +74,10 @@ int main(int argc, char** argv) {
   hipDeviceProp_t prop;
   HIP_CHECK(hipGetDeviceProperties(&prop, device));
 
+  const char* gpu_name =
+      prop.name[0] ? prop.name : "AMD GPU";  // some boxes report no name
   std::printf("mi-stream: %s (%s), %d visible GPU(s), %d CUs, %.0f GiB VRAM\n",
-              prop.name, prop.gcnArchName, ndev, prop.multiProcessorCount,
+              gpu_name, prop.gcnArchName, ndev, prop.multiProcessorCount,
               (double)prop.totalGlobalMem / (1 << 30));
 
   const int64_t n = mib * (1 << 20) / 4;  // fp32 elements
@@ -90,6 +96,46 @@ int main(int argc, char** argv) {
   dim3 grid((uint32_t)k3samd_kern::stream_grid(n4));
   dim3 block(k3samd_kern::kThreadsPerBlock);
   const float s = 2.5f;
+
+  if (tune) {
+    // sweep block-size x unroll x grid-occupancy for the grid-stride
+    // non-temporal triad; prints GB/s per config (exploration tool)
+    const double bytes = 3 * buf_bytes;
+    auto time_one = [&](auto kern, int tpb, int blocks) -> double {
+      hipLaunchKernelGGL(kern, dim3(blocks), dim3(tpb), 0, 0, a, b, c, s, n4);
+      hipLaunchKernelGGL(kern, dim3(blocks), dim3(tpb), 0, 0, a, b, c, s, n4);
+      (void)hipDeviceSynchronize();
+      float best = 1e30f;
+      for (int it = 0; it < iters; ++it) {
+        (void)hipEventRecord(ev0);
+        hipLaunchKernelGGL(kern, dim3(blocks), dim3(tpb), 0, 0, a, b, c, s, n4);
+        (void)hipEventRecord(ev1);
+        (void)hipEventSynchronize(ev1);
+        float ms;
+        (void)hipEventElapsedTime(&ms, ev0, ev1);
+        if (ms < best) best = ms;
+      }
+      return gbps(bytes, best);
+    };
+    std::printf("tune: grid-stride nt triad, %lld MiB buffers\n",
+                (long long)mib);
+    std::printf("%8s %6s %8s %10s\n", "tpb", "unroll", "blocks", "GB/s");
+    for (int tpb : {256, 512, 1024}) {
+      for (int wavesper : {4, 8, 16, 32}) {  // blocks = CUs*waves*64/tpb
+        int blocks = 256 * wavesper * 64 / tpb;
+        double g1 = time_one(k3samd_kern::stream_triad_gs_kernel<true, 1>,
+                             tpb, blocks);
+        double g2 = time_one(k3samd_kern::stream_triad_gs_kernel<true, 2>,
+                             tpb, blocks);
+        double g4 = time_one(k3samd_kern::stream_triad_gs_kernel<true, 4>,
+                             tpb, blocks);
+        std::printf("%8d %6d %8d  U1 %8.1f U2 %8.1f U4 %8.1f\n", tpb, 1,
+                    blocks, g1, g2, g4);
+      }
+    }
+    (void)hipFree(a); (void)hipFree(b); (void)hipFree(c);
+    return 0;
+  }
 
   struct Row {
     const char* name;
