@@ -224,7 +224,7 @@ int main(int argc, char** argv) {
       "{\"payload\": \"mi-stream\", \"gpu\": \"%s\", \"arch\": \"%s\", "
       "\"buffer_MiB\": %lld, \"triad_gbps\": %.1f, \"mfma_bf16_tflops\": "
       "%.1f}\n",
-      prop.name, prop.gcnArchName, (long long)mib, triad_best, mfma_tf);
+      gpu_name, prop.gcnArchName, (long long)mib, triad_best, mfma_tf);
 
   HIP_CHECK(hipFree(a));
   HIP_CHECK(hipFree(b));

@@ -31,6 +31,14 @@ def test_selftest_passes():
     assert "all checks passed" in proc.stdout
 
 
+def test_selftest_asan():
+    """Memory-safety pass over the parsers (HPACK/yaml/json/proto)."""
+    proc = subprocess.run(["make", "-C", str(REPO / "native"), "asan"],
+                          capture_output=True, text=True, timeout=300)
+    assert proc.returncode == 0, proc.stdout + proc.stderr
+    assert "all checks passed" in proc.stdout
+
+
 def test_tsan_grpc_exercise(tmp_path):
     """Drive the TSan build of the plugin through concurrent RPCs (options,
     allocate, a live ListAndWatch stream + health flips) and require zero
