@@ -179,15 +179,21 @@ void HpackDecoder::evict() {
 }
 
 bool HpackDecoder::decode(std::string_view block, std::vector<Header>& out) {
+  // header-list size guard (HTTP/2 SETTINGS_MAX_HEADER_LIST_SIZE analog):
+  // a malicious peer must not expand a small block into unbounded memory
+  constexpr size_t kMaxHeaderListBytes = 1 << 20;
+  size_t total = 0;
   BitReader r{(const uint8_t*)block.data(),
               (const uint8_t*)block.data() + block.size()};
   while (r.p < r.end) {
+    if (total > kMaxHeaderListBytes) return false;
     uint8_t b = *r.p;
     if (b & 0x80) {  // indexed header field
       uint64_t idx;
       if (!r.read_int(7, idx)) return false;
       Header h;
       if (!lookup(idx, h)) return false;
+      total += h.first.size() + h.second.size() + 32;
       out.push_back(h);
     } else if (b & 0x40) {  // literal with incremental indexing
       uint64_t idx;
@@ -200,6 +206,7 @@ bool HpackDecoder::decode(std::string_view block, std::vector<Header>& out) {
       }
       if (!r.read_string(h.second)) return false;
       add_dynamic(h.first, h.second);
+      total += h.first.size() + h.second.size() + 32;
       out.push_back(h);
     } else if (b & 0x20) {  // dynamic table size update
       uint64_t sz;
@@ -217,6 +224,7 @@ bool HpackDecoder::decode(std::string_view block, std::vector<Header>& out) {
         return false;
       }
       if (!r.read_string(h.second)) return false;
+      total += h.first.size() + h.second.size() + 32;
       out.push_back(h);
     }
   }

@@ -124,10 +124,21 @@ int main(int argc, char** argv) {
     HIP_CHECK(hipEventDestroy(ev1));
   }
   std::printf("+-----------+------------+------------+\n");
+  // xGMI sanity verdict: on the fully-connected MI355X node
+  // a healthy multi-GPU all-reduce must beat ONE xGMI link's ~153 GB/s —
+  // proof that RCCL drives parallel p2p paths through the injected
+  // /dev/dri nodes rather than a degenerate single-link/host path.
+  constexpr double kXgmiLinkGbps = 153.0;
+  bool xgmi_ok = true;
+  if (ngpus > 1) {
+    xgmi_ok = last_busbw > kXgmiLinkGbps;
+    std::printf("xGMI p2p check (busbw %.1f GB/s vs single link %.0f): %s\n",
+                last_busbw, kXgmiLinkGbps, xgmi_ok ? "PASS" : "FAIL");
+  }
   std::printf(
       "{\"payload\": \"mi-allreduce\", \"n_gpus\": %d, \"max_busbw_gbps\": "
-      "%.1f}\n",
-      ngpus, last_busbw);
+      "%.1f, \"xgmi_p2p_ok\": %s}\n",
+      ngpus, last_busbw, xgmi_ok ? "true" : "false");
 
   for (int i = 0; i < ngpus; ++i) {
     ncclCommDestroy(comms[i]);
