@@ -42,6 +42,7 @@ int main(int argc, char** argv) {
   int health_poll_ms = 5000;
   bool oneshot = false;
   bool use_cdi = false;
+  std::string metrics_addr;
 
   for (int i = 1; i < argc; ++i) {
     auto arg = [&](const char* name) -> const char* {
@@ -54,6 +55,7 @@ int main(int argc, char** argv) {
     else if (const char* v = arg("--health-poll-ms")) health_poll_ms = std::atoi(v);
     else if (!std::strcmp(argv[i], "--no-register")) kubelet_sock.clear();
     else if (!std::strcmp(argv[i], "--use-cdi")) use_cdi = true;
+    else if (const char* v = arg("--metrics-addr")) metrics_addr = v;
     else if (!std::strcmp(argv[i], "--oneshot")) oneshot = true;
     else if (!std::strcmp(argv[i], "--help") || !std::strcmp(argv[i], "-h")) {
       std::printf("k3samd-device-plugin [--config F] [--plugin-sock S] "
@@ -102,7 +104,8 @@ int main(int argc, char** argv) {
   std::signal(SIGINT, on_signal);
   std::signal(SIGTERM, on_signal);
 
-  if (!plugin.serve(plugin_sock, kubelet_sock, health_poll_ms)) return 1;
+  if (!plugin.serve(plugin_sock, kubelet_sock, health_poll_ms, metrics_addr))
+    return 1;
   std::fprintf(stderr,
                "k3samd-device-plugin: serving %s on %s (%zu devices)%s\n",
                plugin.advertised_resource().c_str(), plugin_sock.c_str(),

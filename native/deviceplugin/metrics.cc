@@ -1,0 +1,92 @@
+#include "metrics.h"
+
+#include <arpa/inet.h>
+#include <cstring>
+#include <netinet/in.h>
+#include <sys/socket.h>
+#include <sys/un.h>
+#include <unistd.h>
+
+namespace k3samd {
+
+MetricsServer::~MetricsServer() { stop(); }
+
+bool MetricsServer::start(const std::string& addr, RenderFn render) {
+  render_ = std::move(render);
+  int fd = -1;
+  if (addr.rfind("unix:", 0) == 0) {
+    std::string path = addr.substr(5);
+    ::unlink(path.c_str());
+    fd = ::socket(AF_UNIX, SOCK_STREAM | SOCK_CLOEXEC, 0);
+    if (fd < 0) return false;
+    sockaddr_un sa{};
+    sa.sun_family = AF_UNIX;
+    if (path.size() >= sizeof(sa.sun_path)) {
+      ::close(fd);
+      return false;
+    }
+    std::strcpy(sa.sun_path, path.c_str());
+    if (::bind(fd, (sockaddr*)&sa, sizeof(sa)) != 0 || ::listen(fd, 8) != 0) {
+      ::close(fd);
+      return false;
+    }
+  } else {
+    size_t colon = addr.find_last_of(':');
+    if (colon == std::string::npos) return false;
+    std::string host = addr.substr(0, colon);
+    int port = std::atoi(addr.c_str() + colon + 1);
+    fd = ::socket(AF_INET, SOCK_STREAM | SOCK_CLOEXEC, 0);
+    if (fd < 0) return false;
+    int one = 1;
+    ::setsockopt(fd, SOL_SOCKET, SO_REUSEADDR, &one, sizeof(one));
+    sockaddr_in sa{};
+    sa.sin_family = AF_INET;
+    sa.sin_port = htons((uint16_t)port);
+    if (::inet_pton(AF_INET, host.c_str(), &sa.sin_addr) != 1) {
+      ::close(fd);
+      return false;
+    }
+    if (::bind(fd, (sockaddr*)&sa, sizeof(sa)) != 0 || ::listen(fd, 8) != 0) {
+      ::close(fd);
+      return false;
+    }
+  }
+  listen_fd_.store(fd);
+  stopping_.store(false);
+  thread_ = std::thread([this] { serve_loop(); });
+  return true;
+}
+
+void MetricsServer::stop() {
+  stopping_.store(true);
+  int fd = listen_fd_.exchange(-1);
+  if (fd >= 0) {
+    ::shutdown(fd, SHUT_RDWR);
+    ::close(fd);
+  }
+  if (thread_.joinable()) thread_.join();
+}
+
+void MetricsServer::serve_loop() {
+  while (!stopping_.load()) {
+    int lfd = listen_fd_.load();
+    if (lfd < 0) break;
+    int cfd = ::accept(lfd, nullptr, nullptr);
+    if (cfd < 0) break;
+    // read whatever request line arrives (we answer every request the same)
+    char buf[1024];
+    (void)::read(cfd, buf, sizeof(buf));
+    std::string body = render_ ? render_() : "";
+    char head[160];
+    int n = std::snprintf(head, sizeof(head),
+                          "HTTP/1.0 200 OK\r\n"
+                          "Content-Type: text/plain; version=0.0.4\r\n"
+                          "Content-Length: %zu\r\n\r\n",
+                          body.size());
+    (void)::write(cfd, head, (size_t)n);
+    (void)::write(cfd, body.data(), body.size());
+    ::close(cfd);
+  }
+}
+
+}  // namespace k3samd

@@ -138,6 +138,7 @@ GrpcStatus DevicePlugin::handle_allocate(const std::string& req,
   for (const auto& ids : containers) {
     if (cfg_.replicas > 1 && cfg_.fail_requests_greater_than_one &&
         ids.size() > 1) {
+      metrics_.allocation_errors_total.fetch_add(1);
       return {3,
               "request for more than one " + advertised_resource() +
                   " is not allowed with time-slicing "
@@ -152,7 +153,10 @@ GrpcStatus DevicePlugin::handle_allocate(const std::string& req,
           gi = vd.gpu_index;
           break;
         }
-      if (gi < 0) return {3, "unknown device id " + id};
+      if (gi < 0) {
+        metrics_.allocation_errors_total.fetch_add(1);
+        return {3, "unknown device id " + id};
+      }
       gpus.insert(gi);
     }
     std::string visible, minors;
@@ -185,6 +189,8 @@ GrpcStatus DevicePlugin::handle_allocate(const std::string& req,
     std::fprintf(stderr,
                  "deviceplugin: Allocate %zu id(s) -> gpus [%s] minors [%s]\n",
                  ids.size(), visible.c_str(), minors.c_str());
+    metrics_.allocations_total.fetch_add(1);
+    metrics_.allocated_devices_total.fetch_add(gpus.size());
     crs.push_back(std::move(cr));
   }
   resp = dp::encode_allocate_response(crs);
@@ -237,6 +243,7 @@ bool DevicePlugin::poll_health_once() {
     bool healthy = present.count(g.stable_id()) > 0;
     if (healthy != vd.healthy) {
       vd.healthy = healthy;
+      metrics_.health_transitions_total.fetch_add(1);
       changed = true;
     }
   }
@@ -253,6 +260,7 @@ GrpcStatus DevicePlugin::handle_list_and_watch(
     std::lock_guard<std::mutex> lk(mu_);
     seen_gen = generation_;
   }
+  metrics_.list_and_watch_updates_total.fetch_add(1);
   if (!write(dp::encode_list_and_watch(current_device_list())))
     return GrpcStatus::Ok();
   // Push an update whenever the device list changes (kubelet keeps this
@@ -267,13 +275,50 @@ GrpcStatus DevicePlugin::handle_list_and_watch(
     }
     if (gen == seen_gen) continue;
     seen_gen = gen;
+    metrics_.list_and_watch_updates_total.fetch_add(1);
     if (!write(dp::encode_list_and_watch(current_device_list()))) break;
   }
   return GrpcStatus::Ok();
 }
 
+std::string DevicePlugin::render_metrics() {
+  size_t healthy = 0, unhealthy = 0;
+  {
+    std::lock_guard<std::mutex> lk(mu_);
+    for (const auto& vd : devices_) (vd.healthy ? healthy : unhealthy)++;
+  }
+  char buf[2048];
+  std::snprintf(
+      buf, sizeof(buf),
+      "# HELP k3samd_gpu_devices Advertised devices by health.\n"
+      "# TYPE k3samd_gpu_devices gauge\n"
+      "k3samd_gpu_devices{health=\"healthy\"} %zu\n"
+      "k3samd_gpu_devices{health=\"unhealthy\"} %zu\n"
+      "# TYPE k3samd_allocations_total counter\n"
+      "k3samd_allocations_total %llu\n"
+      "# TYPE k3samd_allocated_devices_total counter\n"
+      "k3samd_allocated_devices_total %llu\n"
+      "# TYPE k3samd_allocation_errors_total counter\n"
+      "k3samd_allocation_errors_total %llu\n"
+      "# TYPE k3samd_list_and_watch_updates_total counter\n"
+      "k3samd_list_and_watch_updates_total %llu\n"
+      "# TYPE k3samd_registrations_total counter\n"
+      "k3samd_registrations_total %llu\n"
+      "# TYPE k3samd_health_transitions_total counter\n"
+      "k3samd_health_transitions_total %llu\n",
+      healthy, unhealthy,
+      (unsigned long long)metrics_.allocations_total.load(),
+      (unsigned long long)metrics_.allocated_devices_total.load(),
+      (unsigned long long)metrics_.allocation_errors_total.load(),
+      (unsigned long long)metrics_.list_and_watch_updates_total.load(),
+      (unsigned long long)metrics_.registrations_total.load(),
+      (unsigned long long)metrics_.health_transitions_total.load());
+  return buf;
+}
+
 bool DevicePlugin::serve(const std::string& plugin_sock,
-                         const std::string& kubelet_sock, int health_poll_ms) {
+                         const std::string& kubelet_sock, int health_poll_ms,
+                         const std::string& metrics_addr) {
   server_.add_unary(dp::kOptionsPath,
                     [this](const std::string&, std::string& resp) {
                       resp = handle_options();
@@ -341,6 +386,12 @@ bool DevicePlugin::serve(const std::string& plugin_sock,
     });
   }
 
+  if (!metrics_addr.empty() &&
+      !metrics_server_.start(metrics_addr, [this] { return render_metrics(); })) {
+    std::fprintf(stderr, "deviceplugin: cannot bind metrics %s\n",
+                 metrics_addr.c_str());
+  }
+
   if (health_poll_ms > 0) {
     health_thread_ = std::thread([this, health_poll_ms] {
       while (!stopping_.load()) {
@@ -360,6 +411,8 @@ bool DevicePlugin::register_with_kubelet() {
                                            : plugin_sock_.substr(slash + 1);
   rr.resource_name = advertised_resource();
   auto res = grpc_unary_call(kubelet_sock_, dp::kRegisterPath, rr.encode());
+  if (res.transport_ok && res.grpc_status == 0)
+    metrics_.registrations_total.fetch_add(1);
   if (!res.transport_ok || res.grpc_status != 0) {
     std::fprintf(stderr,
                  "deviceplugin: Register with kubelet failed: %s (grpc=%d %s)\n",
@@ -373,6 +426,7 @@ void DevicePlugin::stop() {
   stopping_.store(true);
   if (health_thread_.joinable()) health_thread_.join();
   if (reregister_thread_.joinable()) reregister_thread_.join();
+  metrics_server_.stop();
   server_.stop();
 }
 

@@ -17,6 +17,7 @@
 #include "../grpc/grpc_transport.h"
 #include "../topology/kfd_topology.h"
 #include "dp_messages.h"
+#include "metrics.h"
 
 namespace k3samd {
 
@@ -58,9 +59,12 @@ class DevicePlugin {
 
   // Serve on `plugin_sock` and (if kubelet_sock non-empty) register with
   // kubelet. Non-blocking; returns false if the socket can't be bound or
-  // registration fails.
+  // registration fails. `metrics_addr` ("host:port" or "unix:/path", empty
+  // = disabled) exposes Prometheus metrics.
   bool serve(const std::string& plugin_sock, const std::string& kubelet_sock,
-             int health_poll_ms = 5000);
+             int health_poll_ms = 5000,
+             const std::string& metrics_addr = "");
+  std::string render_metrics();
   void stop();
 
   // --- logic, exposed for unit tests ---
@@ -90,6 +94,8 @@ class DevicePlugin {
   bool register_with_kubelet();
 
   GrpcServer server_;
+  PluginMetrics metrics_;
+  MetricsServer metrics_server_;
   std::thread health_thread_;
   std::thread reregister_thread_;
   std::string plugin_sock_;

@@ -1,0 +1,93 @@
+"""Prometheus metrics endpoint tests (device plugin --metrics-addr)."""
+
+import socket
+import subprocess
+import time
+from pathlib import Path
+
+import pytest
+
+import pb_v1beta1 as pb
+from sysfs_builder import build_tree
+
+REPO = Path(__file__).resolve().parent.parent
+PLUGIN = REPO / "native" / "bin" / "k3samd-device-plugin"
+
+IDENT = lambda b: b  # noqa: E731
+
+
+@pytest.fixture(scope="session", autouse=True)
+def build_native():
+    subprocess.run(["make", "-C", str(REPO / "native"), "-j8"], check=True,
+                   capture_output=True)
+
+
+def scrape(metrics_sock):
+    s = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+    s.connect(metrics_sock)
+    s.sendall(b"GET /metrics HTTP/1.0\r\n\r\n")
+    data = b""
+    while chunk := s.recv(65536):
+        data += chunk
+    s.close()
+    head, body = data.split(b"\r\n\r\n", 1)
+    assert b"200 OK" in head
+    return None, body.decode()
+
+
+def parse_metrics(body: str) -> dict:
+    out = {}
+    for line in body.splitlines():
+        if line.startswith("#") or not line.strip():
+            continue
+        k, v = line.rsplit(" ", 1)
+        out[k] = float(v)
+    return out
+
+
+def test_metrics_endpoint(tmp_path):
+    root = build_tree(tmp_path / "sys", n_gpus=2)
+    sock = str(tmp_path / "amd.sock")
+    msock = str(tmp_path / "metrics.sock")
+    proc = subprocess.Popen(
+        [str(PLUGIN), "--plugin-sock", sock, "--no-register",
+         "--metrics-addr", f"unix:{msock}", "--health-poll-ms", "0"],
+        env={"K3SAMD_SYSFS_ROOT": str(root)},
+        stdout=subprocess.PIPE, stderr=subprocess.PIPE)
+    try:
+        deadline = time.time() + 10
+        while not (Path(sock).exists() and Path(msock).exists()):
+            assert time.time() < deadline and proc.poll() is None
+            time.sleep(0.05)
+        _, body = scrape(msock)
+        m = parse_metrics(body)
+        assert m['k3samd_gpu_devices{health="healthy"}'] == 2
+        assert m["k3samd_allocations_total"] == 0
+
+        # drive one allocation through grpc and re-scrape
+        import grpc
+        ch = grpc.insecure_channel(f"unix:{sock}")
+        law = ch.unary_stream("/v1beta1.DevicePlugin/ListAndWatch",
+                              request_serializer=IDENT,
+                              response_deserializer=IDENT)(b"", timeout=10)
+        devs = pb.decode_list_and_watch(next(law))
+        alloc = ch.unary_unary("/v1beta1.DevicePlugin/Allocate",
+                               request_serializer=IDENT,
+                               response_deserializer=IDENT)
+        alloc(pb.encode_allocate_request([[devs[0]["id"], devs[1]["id"]]]),
+              timeout=10)
+        try:
+            alloc(pb.encode_allocate_request([["bogus"]]), timeout=10)
+        except grpc.RpcError:
+            pass
+        ch.close()
+
+        _, body = scrape(msock)
+        m = parse_metrics(body)
+        assert m["k3samd_allocations_total"] == 1
+        assert m["k3samd_allocated_devices_total"] == 2
+        assert m["k3samd_allocation_errors_total"] == 1
+        assert m["k3samd_list_and_watch_updates_total"] >= 1
+    finally:
+        proc.terminate()
+        proc.wait(timeout=10)
