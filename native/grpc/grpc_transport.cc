@@ -1,5 +1,6 @@
 #include "grpc_transport.h"
 
+#include <chrono>
 #include <cstring>
 #include <condition_variable>
 #include <poll.h>
@@ -145,7 +146,13 @@ struct H2Conn : std::enable_shared_from_this<H2Conn> {
   std::atomic<int> send_window{kDefaultWindow};
   int peer_initial_window = kDefaultWindow;
 
-  ~H2Conn() { close_fd(); }
+  ~H2Conn() {
+    close_fd();
+    if (fd >= 0) {
+      ::close(fd);
+      fd = -1;
+    }
+  }
 
   void close_fd() {
     bool was = closed.exchange(true);
@@ -255,25 +262,43 @@ void GrpcServer::stop() {
     for (auto& c : conns_) c->close_fd();
   }
   if (accept_thread_.joinable()) accept_thread_.join();
-  // join every connection/stream thread (their fds are shut down above, so
-  // they exit promptly; new stream threads can only be added by live
-  // connection threads, which this loop drains)
+  // join every connection/stream thread (their fds are shut down above,
+  // so they exit promptly; new stream threads can only be spawned by
+  // connection threads this loop is joining)
   for (;;) {
-    std::thread t;
+    Tracked tr;
     {
       std::lock_guard<std::mutex> lk(threads_mu_);
       if (threads_.empty()) break;
-      t = std::move(threads_.back());
+      tr = std::move(threads_.back());
       threads_.pop_back();
     }
-    if (t.joinable()) t.join();
+    if (tr.t.joinable()) tr.t.join();
   }
   running_.store(false);
 }
 
-void GrpcServer::track_thread(std::thread t) {
+void GrpcServer::reap_locked() {
+  for (size_t i = 0; i < threads_.size();) {
+    if (threads_[i].done->load()) {
+      if (threads_[i].t.joinable()) threads_[i].t.join();  // finishes now
+      threads_[i] = std::move(threads_.back());
+      threads_.pop_back();
+    } else {
+      ++i;
+    }
+  }
+}
+
+void GrpcServer::spawn(std::function<void()> fn) {
+  auto done = std::make_shared<std::atomic<bool>>(false);
+  std::thread t([fn = std::move(fn), done] {
+    fn();
+    done->store(true);
+  });
   std::lock_guard<std::mutex> lk(threads_mu_);
-  threads_.push_back(std::move(t));
+  reap_locked();
+  threads_.push_back({std::move(t), std::move(done)});
 }
 
 namespace {
@@ -322,7 +347,7 @@ void GrpcServer::serve_loop() {
       std::lock_guard<std::mutex> lk(conns_mu_);
       conns_.insert(conn);
     }
-    track_thread(std::thread([this, conn] {
+    spawn([this, conn] {
       // --- connection handshake ---
       char preface[kPrefaceLen];
       if (!read_full(conn->fd, preface, kPrefaceLen, 10000) ||
@@ -502,7 +527,7 @@ void GrpcServer::serve_loop() {
             auto sit = stream_.find(path);
             if (sit != stream_.end()) {
               StreamHandler handler = sit->second;
-              track_thread(std::thread([conn, st, handler, req] {
+              spawn([conn, st, handler, req] {
                 send_response_headers(conn, st);
                 auto write = [conn, st](const std::string& msg) -> bool {
                   if (conn->closed.load() || st->cancelled.load()) return false;
@@ -511,7 +536,7 @@ void GrpcServer::serve_loop() {
                 GrpcStatus gs = handler(req, write);
                 if (!conn->closed.load() && !st->cancelled.load())
                   send_trailers(conn, st, gs, false);
-              }));
+              });
               continue;
             }
             send_trailers(conn, st, {12, "unknown method " + path}, true);
@@ -524,7 +549,7 @@ void GrpcServer::serve_loop() {
         std::lock_guard<std::mutex> lk(conns_mu_);
         conns_.erase(conn);
       }
-    }));
+    });
   }
 }
 

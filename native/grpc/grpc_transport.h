@@ -13,6 +13,7 @@
 #pragma once
 
 #include <atomic>
+#include <condition_variable>
 #include <functional>
 #include <map>
 #include <memory>
@@ -65,11 +66,18 @@ class GrpcServer {
   std::atomic<bool> stopping_{false};
   std::mutex conns_mu_;
   std::set<std::shared_ptr<struct H2Conn>> conns_;
-  // every spawned thread (connection readers + stream handlers) is tracked
-  // and joined in stop() so no thread outlives the server object
-  void track_thread(std::thread t);
+  // Every spawned thread (connection readers + stream handlers) stays
+  // joinable (detached threads recycle TSan thread slots and produce
+  // false reports) and is reaped: each marks a done flag on exit, spawn()
+  // joins finished ones, stop() joins everything.
+  struct Tracked {
+    std::thread t;
+    std::shared_ptr<std::atomic<bool>> done;
+  };
+  void spawn(std::function<void()> fn);
+  void reap_locked();
   std::mutex threads_mu_;
-  std::vector<std::thread> threads_;
+  std::vector<Tracked> threads_;
 };
 
 struct UnaryCallResult {
