@@ -449,7 +449,14 @@ void GrpcServer::serve_loop() {
                 if (it != conn->streams.end()) st = it->second;
               }
               if (st) {
-                st->data.append(frag);
+                // bound request buffering (our RPCs are tiny; a peer must
+                // not be able to balloon server memory)
+                constexpr size_t kMaxRequestBytes = 16u << 20;
+                if (st->data.size() + frag.size() > kMaxRequestBytes) {
+                  st->cancelled.store(true);
+                } else {
+                  st->data.append(frag);
+                }
                 if (f.flags & kEndStream) st->request_complete = true;
               }
               // replenish peer's send window (conn + stream)
