@@ -35,7 +35,9 @@ def _needs_rebuild() -> bool:
     so = _ext_path()
     if not so.exists():
         return True
-    return HIP_SRC.stat().st_mtime > so.stat().st_mtime
+    so_mtime = so.stat().st_mtime
+    deps = [HIP_SRC, HIP_SRC.parent / "stream_kernels.h"]
+    return any(d.exists() and d.stat().st_mtime > so_mtime for d in deps)
 
 
 def build_extension(verbose: bool = True) -> Path:
