@@ -243,6 +243,7 @@ bool GrpcServer::start(const std::string& unix_path) {
     return false;
   }
   listen_fd_.store(fd);
+  bound_path_ = unix_path;
   running_.store(true);
   stopping_.store(false);
   accept_thread_ = std::thread([this] { serve_loop(); });
@@ -256,6 +257,7 @@ void GrpcServer::stop() {
   if (fd >= 0) {
     ::shutdown(fd, SHUT_RDWR);
     ::close(fd);
+    if (!bound_path_.empty()) ::unlink(bound_path_.c_str());
   }
   {
     std::lock_guard<std::mutex> lk(conns_mu_);

@@ -61,6 +61,7 @@ class GrpcServer {
   std::map<std::string, UnaryHandler> unary_;
   std::map<std::string, StreamHandler> stream_;
   std::atomic<int> listen_fd_{-1};
+  std::string bound_path_;
   std::thread accept_thread_;
   std::atomic<bool> running_{false};
   std::atomic<bool> stopping_{false};

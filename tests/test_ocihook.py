@@ -152,6 +152,45 @@ def test_cpu_only_node_injects_nothing(tmp_path):
     assert "/dev/kfd" not in json.dumps(out)
 
 
+def test_runc_passthrough_create(tmp_path):
+    """Full runc-wrapper path: `create --bundle B id` transforms the
+    bundle's config.json, then execs the (stubbed) real runc with
+    identical arguments."""
+    sysfs = build_tree(tmp_path / "sys", n_gpus=1)
+    bundle = tmp_path / "bundle"
+    bundle.mkdir()
+    (bundle / "config.json").write_text(json.dumps(BASE_SPEC))
+    # stub runc that records its argv
+    stub = tmp_path / "runc-stub"
+    argv_log = tmp_path / "argv.txt"
+    stub.write_text(f"#!/bin/sh\necho \"$@\" > {argv_log}\nexit 0\n")
+    stub.chmod(0o755)
+    proc = subprocess.run(
+        [str(RUNTIME), "--root", "/run/x", "create",
+         "--bundle", str(bundle), "ctr-1"],
+        env={"K3SAMD_SYSFS_ROOT": str(sysfs),
+             "K3SAMD_DEV_ROOT": str(tmp_path / "nodev"),
+             "K3SAMD_RUNC_PATH": str(stub)},
+        capture_output=True, text=True, timeout=60)
+    assert proc.returncode == 0, proc.stderr
+    assert argv_log.read_text().split() == [
+        "--root", "/run/x", "create", "--bundle", str(bundle), "ctr-1"]
+    out = json.loads((bundle / "config.json").read_text())
+    assert "/dev/kfd" in device_paths(out)
+
+
+def test_runc_passthrough_noncreate_untouched(tmp_path):
+    """`state`/`delete` etc. must pass through without touching configs."""
+    stub = tmp_path / "runc-stub"
+    stub.write_text("#!/bin/sh\nexit 0\n")
+    stub.chmod(0o755)
+    proc = subprocess.run(
+        [str(RUNTIME), "state", "ctr-1"],
+        env={"K3SAMD_RUNC_PATH": str(stub)},
+        capture_output=True, text=True, timeout=60)
+    assert proc.returncode == 0, proc.stderr
+
+
 def test_malformed_config_fails(tmp_path):
     cfg = tmp_path / "config.json"
     cfg.write_text("{not json")
