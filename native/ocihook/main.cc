@@ -83,15 +83,27 @@ int main(int argc, char** argv) {
     return transform_config(argv[2]) ? 0 : 1;
   }
 
-  // runc-compatible invocation: find subcommand + bundle
+  // runc-compatible invocation: find subcommand + bundle. Global flags can
+  // take values ("--root /run/..."), so the subcommand is the first token
+  // matching runc's command set, not merely the first non-dash token.
+  static const char* kRuncCommands[] = {
+      "checkpoint", "create", "delete", "events", "exec", "features", "init",
+      "kill", "list", "pause", "ps", "restore", "resume", "run", "spec",
+      "start", "state", "update"};
   const char* subcommand = nullptr;
   std::string bundle = ".";
   for (int i = 1; i < argc; ++i) {
     const char* a = argv[i];
-    if (a[0] != '-' && !subcommand) {
-      subcommand = a;
-    } else if ((!std::strcmp(a, "--bundle") || !std::strcmp(a, "-b")) &&
-               i + 1 < argc) {
+    if (!subcommand && a[0] != '-') {
+      for (const char* c : kRuncCommands) {
+        if (!std::strcmp(a, c)) {
+          subcommand = a;
+          break;
+        }
+      }
+    }
+    if ((!std::strcmp(a, "--bundle") || !std::strcmp(a, "-b")) &&
+        i + 1 < argc) {
       bundle = argv[++i];
     } else if (!std::strncmp(a, "--bundle=", 9)) {
       bundle = a + 9;
