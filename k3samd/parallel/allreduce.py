@@ -17,37 +17,65 @@ import os
 import time
 
 
-def busbw_sweep(sizes_bytes, iters=20, device=None, group=None):
-    """All-reduce each size `iters` times; returns [{bytes, algbw, busbw}].
+# busbw normalization factors (bytes actually crossing links per byte of
+# payload, ring-equivalent — the standard nccl-tests convention)
+_BUSBW_FACTOR = {
+    "all_reduce": lambda n: 2 * (n - 1) / n,
+    "all_gather": lambda n: (n - 1) / n,
+    "reduce_scatter": lambda n: (n - 1) / n,
+    "broadcast": lambda n: 1.0,
+}
 
-    busbw = 2*(n-1)/n * algbw — the standard ring-equivalent bus bandwidth
-    normalization, comparable across world sizes.
-    """
+
+def _run_collective(op, t, world, device, group):
+    import torch
+    import torch.distributed as dist
+
+    if op == "all_reduce":
+        dist.all_reduce(t, group=group)
+    elif op == "broadcast":
+        dist.broadcast(t, src=0, group=group)
+    elif op == "all_gather":
+        out = torch.empty(t.numel() * world, dtype=t.dtype, device=t.device)
+        dist.all_gather_into_tensor(out, t, group=group)
+    elif op == "reduce_scatter":
+        inp = torch.empty(t.numel() * world, dtype=t.dtype, device=t.device)
+        dist.reduce_scatter_tensor(t, inp, group=group)
+    else:
+        raise ValueError(f"unknown op {op}")
+
+
+def busbw_sweep(sizes_bytes, iters=20, device=None, group=None,
+                op="all_reduce"):
+    """Run `op` over each size `iters` times; returns
+    [{bytes, algbw, busbw}]. busbw uses the standard ring-equivalent
+    normalization so values are comparable across world sizes and ops."""
     import torch
     import torch.distributed as dist
 
     world = dist.get_world_size(group)
+    factor = _BUSBW_FACTOR[op](world)
     rows = []
     for bytes_ in sizes_bytes:
         n = max(int(bytes_) // 4, 1)
         t = torch.rand(n, dtype=torch.float32, device=device)
         for _ in range(3):
-            dist.all_reduce(t, group=group)
+            _run_collective(op, t, world, device, group)
         if device is not None and str(device).startswith("cuda"):
             torch.cuda.synchronize(device)
         dist.barrier(group)
         t0 = time.perf_counter()
         for _ in range(iters):
-            dist.all_reduce(t, group=group)
+            _run_collective(op, t, world, device, group)
         if device is not None and str(device).startswith("cuda"):
             torch.cuda.synchronize(device)
         elapsed = time.perf_counter() - t0
         # max over ranks so the report reflects the slowest participant
-        e = torch.tensor([elapsed], dtype=torch.float64)
+        e = torch.tensor([elapsed], dtype=torch.float64, device=device)
         dist.all_reduce(e, op=dist.ReduceOp.MAX, group=group)
         per_op = float(e[0]) / iters
         algbw = n * 4 / per_op / 1e9
-        busbw = algbw * 2 * (world - 1) / world
+        busbw = algbw * factor
         rows.append({"bytes": n * 4, "algbw_gbps": round(algbw, 2),
                      "busbw_gbps": round(busbw, 2)})
     return rows
@@ -62,6 +90,8 @@ def main(argv=None) -> int:
     p.add_argument("--max-mib", type=int, default=1024)
     p.add_argument("--iters", type=int, default=20)
     p.add_argument("--backend", default="auto")
+    p.add_argument("--op", default="all_reduce",
+                   choices=sorted(_BUSBW_FACTOR))
     args = p.parse_args(argv)
 
     backend = args.backend
@@ -84,14 +114,15 @@ def main(argv=None) -> int:
     while b <= args.max_mib << 20:
         sizes.append(b)
         b *= 4
-    rows = busbw_sweep(sizes, iters=args.iters, device=device)
+    rows = busbw_sweep(sizes, iters=args.iters, device=device, op=args.op)
 
     if rank == 0:
         for r in rows:
             print(f"{r['bytes']:>12d} B  algbw {r['algbw_gbps']:8.2f} GB/s"
                   f"  busbw {r['busbw_gbps']:8.2f} GB/s")
         print(json.dumps({"payload": "allreduce", "backend": backend,
-                          "world_size": world, "rows": rows}))
+                          "op": args.op, "world_size": world,
+                          "rows": rows}))
     dist.destroy_process_group()
     return 0
 

@@ -37,6 +37,11 @@ assert all(r["busbw_gbps"] > 0 for r in rows)
 # world=2: busbw = algbw * 2*(2-1)/2 = algbw
 for r in rows:
     assert abs(r["busbw_gbps"] - r["algbw_gbps"]) < 1e-6 + 0.02 * r["algbw_gbps"]
+
+# other collectives run and produce sane numbers on gloo too
+for op in ("all_gather", "broadcast"):
+    rows = busbw_sweep([1 << 14], iters=2, op=op)
+    assert rows[0]["busbw_gbps"] > 0, op
 if rank == 0:
     print("WORKER_OK", json.dumps(rows))
 dist.destroy_process_group()
