@@ -37,6 +37,10 @@ def parse_args(argv=None):
     p.add_argument("--variant", choices=["auto", "plain", "nt"], default="auto",
                    help="plain or non-temporal loads/stores (auto measures both"
                         " during warmup and picks the faster globally)")
+    p.add_argument("--graph", choices=["auto", "on", "off"], default="auto",
+                   help="capture the step in a hipGraph and replay it "
+                        "(amortizes launch overhead; auto = use if capture "
+                        "succeeds)")
     p.add_argument("--scalar", type=float, default=2.5)
     return p.parse_args(argv)
 
@@ -134,9 +138,36 @@ def main(argv=None) -> int:
         use_nt = args.variant == "nt"
     variant = "nt" if use_nt else "plain"
 
+    # ---- optional hipGraph capture of one step ----
+    graph = None
+    if args.graph in ("auto", "on"):
+        try:
+            g = torch.cuda.CUDAGraph()
+            # side-stream warmup required before capture
+            s = torch.cuda.Stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                run_step(use_nt)
+            torch.cuda.current_stream().wait_stream(s)
+            torch.cuda.synchronize()
+            with torch.cuda.graph(g):
+                run_step(use_nt)
+            graph = g
+        except Exception as e:  # pragma: no cover - device dependent
+            if args.graph == "on":
+                raise
+            print(f"# graph capture unavailable, eager launches: {e}",
+                  file=sys.stderr)
+
+    def do_step():
+        if graph is not None:
+            graph.replay()
+        else:
+            run_step(use_nt)
+
     # ---- warmup ----
     for _ in range(args.warmup):
-        run_step(use_nt)
+        do_step()
     torch.cuda.synchronize()
 
     # ---- timed region ----
@@ -144,7 +175,7 @@ def main(argv=None) -> int:
     torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        run_step(use_nt)
+        do_step()
     torch.cuda.synchronize()
     elapsed = time.perf_counter() - t0
     barrier()
@@ -161,6 +192,7 @@ def main(argv=None) -> int:
         res = build_result(value=value, n_gpus=n_gpus, steps=args.steps,
                            warmup=args.warmup, ms_per_step=ms_per_step,
                            buffer_mib=args.buffer_mib, variant=variant)
+        res["config"]["hip_graph"] = graph is not None
         print(json.dumps(res))
 
     if dist is not None:
