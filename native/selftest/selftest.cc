@@ -5,6 +5,8 @@
 
 #include <cstdio>
 #include <cstring>
+#include <iostream>
+#include <iterator>
 #include <string>
 #include <vector>
 
@@ -225,7 +227,52 @@ int test_minijson() {
   return 0;
 }
 
-int main() {
+// --fuzz <mode>: read stdin, feed it to a parser, exit 0 unless the parser
+// crashes/throws-unexpectedly (drives the hypothesis tests in
+// tests/test_fuzz.py; parsers must reject, never die, on malformed input).
+int fuzz_main(const char* mode) {
+  std::string input((std::istreambuf_iterator<char>(std::cin)),
+                    std::istreambuf_iterator<char>());
+  if (!std::strcmp(mode, "hpack")) {
+    k3samd::HpackDecoder dec;
+    std::vector<k3samd::Header> out;
+    (void)dec.decode(input, out);
+    return 0;
+  }
+  if (!std::strcmp(mode, "huffman")) {
+    std::string out;
+    (void)k3samd::hpack_huffman_decode(input, out);
+    return 0;
+  }
+  if (!std::strcmp(mode, "proto")) {
+    std::vector<std::vector<std::string>> a;
+    (void)k3samd::dp::decode_allocate_request(input, a);
+    std::vector<k3samd::dp::PreferredRequest> p;
+    (void)k3samd::dp::decode_preferred_request(input, p);
+    k3samd::dp::RegisterRequest r;
+    (void)k3samd::dp::RegisterRequest::decode(input, r);
+    return 0;
+  }
+  if (!std::strcmp(mode, "yaml")) {
+    try {
+      (void)k3samd::yaml_parse(input);
+    } catch (const std::exception&) {
+    }
+    return 0;
+  }
+  if (!std::strcmp(mode, "json")) {
+    try {
+      (void)k3samd::json_parse(input);
+    } catch (const std::exception&) {
+    }
+    return 0;
+  }
+  std::fprintf(stderr, "unknown fuzz mode %s\n", mode);
+  return 2;
+}
+
+int main(int argc, char** argv) {
+  if (argc >= 3 && !std::strcmp(argv[1], "--fuzz")) return fuzz_main(argv[2]);
   int rc = 0;
   rc |= test_huffman();
   rc |= test_hpack_decode_rfc_c4();

@@ -1,0 +1,66 @@
+"""Property tests: the native parsers must reject, never crash, on
+arbitrary input (they face untrusted peers/config in production)."""
+
+import json
+import subprocess
+from pathlib import Path
+
+import pytest
+from hypothesis import given, settings, strategies as st
+
+REPO = Path(__file__).resolve().parent.parent
+SELFTEST = REPO / "native" / "bin" / "k3samd-selftest"
+
+
+@pytest.fixture(scope="session", autouse=True)
+def build_native():
+    subprocess.run(["make", "-C", str(REPO / "native"), "-j8"], check=True,
+                   capture_output=True)
+
+
+def run_fuzz(mode: str, data: bytes):
+    proc = subprocess.run([str(SELFTEST), "--fuzz", mode], input=data,
+                          capture_output=True, timeout=60)
+    assert proc.returncode == 0, (mode, data[:80], proc.stderr[:200])
+
+
+@settings(max_examples=40, deadline=None)
+@given(st.binary(max_size=512))
+def test_hpack_never_crashes(data):
+    run_fuzz("hpack", data)
+    run_fuzz("huffman", data)
+
+
+@settings(max_examples=40, deadline=None)
+@given(st.binary(max_size=512))
+def test_proto_never_crashes(data):
+    run_fuzz("proto", data)
+
+
+@settings(max_examples=40, deadline=None)
+@given(st.text(max_size=400))
+def test_yaml_never_crashes(data):
+    run_fuzz("yaml", data.encode("utf-8", "replace"))
+
+
+@settings(max_examples=40, deadline=None)
+@given(st.text(max_size=400))
+def test_json_never_crashes(data):
+    run_fuzz("json", data.encode("utf-8", "replace"))
+
+
+# structured JSON round-trips exactly through minijson (via fuzz mode we
+# only get crash coverage; use the selftest for round-trip, plus spot-check
+# python-generated documents parse)
+json_values = st.recursive(
+    st.none() | st.booleans() | st.integers(-2**31, 2**31) |
+    st.text(max_size=20),
+    lambda children: st.lists(children, max_size=4) |
+    st.dictionaries(st.text(max_size=8), children, max_size=4),
+    max_leaves=12)
+
+
+@settings(max_examples=30, deadline=None)
+@given(json_values)
+def test_minijson_accepts_valid_json(value):
+    run_fuzz("json", json.dumps(value).encode())
