@@ -50,9 +50,10 @@ def stream_add(a, b, c, nontemporal=False):
     _require_native().stream_add(a, b, c, nontemporal)
 
 
-def mfma_throughput(out, iters):
-    """Issue `iters` MFMA quads per wave; returns FLOPs issued."""
-    return _require_native().mfma_throughput(out, int(iters))
+def mfma_throughput(out, iters, shape=16):
+    """Issue `iters` MFMA groups per wave (shape 16 = v_mfma_f32_16x16x32,
+    shape 32 = v_mfma_f32_32x32x16); returns FLOPs issued."""
+    return _require_native().mfma_throughput(out, int(iters), int(shape))
 
 
 # Validated on MI355X silicon (first gpurun, 2026-09-13): layout 0 — lane l

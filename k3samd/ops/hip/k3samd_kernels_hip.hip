@@ -116,18 +116,25 @@ void stream_add(torch::Tensor a, torch::Tensor b, torch::Tensor c,
 
 // Launch `out.numel()` blocks x 256 threads, `iters` MFMA quads each.
 // Returns total FLOPs issued so the caller can compute TFLOP/s.
-double mfma_throughput(torch::Tensor out, int64_t iters) {
+double mfma_throughput(torch::Tensor out, int64_t iters, int64_t shape) {
   K3_CHECK(out.is_cuda() && out.scalar_type() == torch::kFloat32 &&
                out.is_contiguous(),
            "out must be contiguous float32 GPU tensor");
+  K3_CHECK(shape == 16 || shape == 32, "shape must be 16 or 32");
   int64_t blocks = out.numel();
   K3_CHECK(blocks > 0 && blocks <= (1 << 22), "bad block count");
   auto stream = at::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL(k3samd_kern::mfma_throughput_kernel, dim3(blocks),
-                     dim3(kThreadsPerBlock), 0, stream, out.data_ptr<float>(),
-                     (int)iters);
+  if (shape == 16) {
+    hipLaunchKernelGGL(k3samd_kern::mfma_throughput_kernel, dim3(blocks),
+                       dim3(kThreadsPerBlock), 0, stream,
+                       out.data_ptr<float>(), (int)iters);
+  } else {
+    hipLaunchKernelGGL(k3samd_kern::mfma_throughput32_kernel, dim3(blocks),
+                       dim3(kThreadsPerBlock), 0, stream,
+                       out.data_ptr<float>(), (int)iters);
+  }
   C10_HIP_KERNEL_LAUNCH_CHECK();
-  // waves/block = 256/64 = 4; 4 accumulators; 16384 FLOP per MFMA
+  // waves/block = 4; (4 acc x 16384 FLOP) or (2 acc x 32768 FLOP) per iter
   return (double)blocks * 4.0 * 4.0 * 16384.0 * (double)iters;
 }
 
@@ -171,7 +178,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("s"), py::arg("nontemporal") = false);
   m.def("stream_add", &stream_add, py::arg("a"), py::arg("b"), py::arg("c"),
         py::arg("nontemporal") = false);
-  m.def("mfma_throughput", &mfma_throughput, py::arg("out"), py::arg("iters"));
+  m.def("mfma_throughput", &mfma_throughput, py::arg("out"), py::arg("iters"),
+        py::arg("shape") = 16);
   m.def("mfma_gemm16", &mfma_gemm16, py::arg("A"), py::arg("B"),
         py::arg("layout") = 0);
   m.def("has_mfma", &has_mfma);

@@ -195,26 +195,31 @@ int main(int argc, char** argv) {
   }
   std::printf("+--------+-------------------+-------------------+\n");
 
-  double mfma_tf = 0;
+  double mfma_tf = 0, mfma32_tf = 0;
   if (do_mfma) {
     const int blocks = 2048, mf_iters = 4096;
     float* out;
     HIP_CHECK(hipMalloc(&out, blocks * sizeof(float)));
-    hipLaunchKernelGGL(k3samd_kern::mfma_throughput_kernel, dim3(blocks),
-                       block, 0, 0, out, 256);  // warm
-    HIP_CHECK(hipDeviceSynchronize());
-    HIP_CHECK(hipEventRecord(ev0));
-    hipLaunchKernelGGL(k3samd_kern::mfma_throughput_kernel, dim3(blocks),
-                       block, 0, 0, out, mf_iters);
-    HIP_CHECK(hipEventRecord(ev1));
-    HIP_CHECK(hipEventSynchronize(ev1));
-    float ms;
-    HIP_CHECK(hipEventElapsedTime(&ms, ev0, ev1));
-    double flops = (double)blocks * 4 * 4 * 16384.0 * mf_iters;
-    mfma_tf = flops / (ms * 1e9);
-    std::printf("| MFMA bf16 (v_mfma_f32_16x16x32_bf16): %8.1f TFLOP/s    |\n",
-                mfma_tf);
+    auto time_mfma = [&](auto kern, double flops_per_wave_iter,
+                         int accs) -> double {
+      hipLaunchKernelGGL(kern, dim3(blocks), block, 0, 0, out, 256);  // warm
+      (void)hipDeviceSynchronize();
+      (void)hipEventRecord(ev0);
+      hipLaunchKernelGGL(kern, dim3(blocks), block, 0, 0, out, mf_iters);
+      (void)hipEventRecord(ev1);
+      (void)hipEventSynchronize(ev1);
+      float ms;
+      (void)hipEventElapsedTime(&ms, ev0, ev1);
+      double flops =
+          (double)blocks * 4 * accs * flops_per_wave_iter * mf_iters;
+      return flops / (ms * 1e9);
+    };
+    mfma_tf = time_mfma(k3samd_kern::mfma_throughput_kernel, 16384.0, 4);
+    mfma32_tf = time_mfma(k3samd_kern::mfma_throughput32_kernel, 32768.0, 2);
+    std::printf("| MFMA bf16 16x16x32: %7.1f TF | 32x32x16: %7.1f TF     |\n",
+                mfma_tf, mfma32_tf);
     std::printf("+--------+-------------------+-------------------+\n");
+    if (mfma32_tf > mfma_tf) mfma_tf = mfma32_tf;
     HIP_CHECK(hipFree(out));
   }
 

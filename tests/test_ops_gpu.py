@@ -85,10 +85,11 @@ def test_mfma_gemm16_vs_torch():
 
 
 @requires_gpu
-def test_mfma_throughput_runs():
+@pytest.mark.parametrize("shape", [16, 32])
+def test_mfma_throughput_runs(shape):
     dev = torch.device("cuda", 0)
     out = torch.zeros(2048, device=dev)
-    flops = ops.mfma_throughput(out, 100)
+    flops = ops.mfma_throughput(out, 100, shape=shape)
     torch.cuda.synchronize()
     assert flops > 0
     assert torch.isfinite(out).all()
