@@ -11,6 +11,7 @@
 //
 // Build: hipcc --offload-arch=gfx950 -O3 mi_stream.hip -o mi-stream
 
+#include <chrono>
 #include <cstdio>
 #include <cstring>
 #include <string>
@@ -46,6 +47,7 @@ int main(int argc, char** argv) {
   int device = 0;
   bool do_mfma = true;
   bool tune = false;
+  bool all_gpus = false;
   for (int i = 1; i < argc; ++i) {
     if (!std::strcmp(argv[i], "--mib") && i + 1 < argc)
       mib = std::atoll(argv[++i]);
@@ -57,9 +59,11 @@ int main(int argc, char** argv) {
       do_mfma = false;
     else if (!std::strcmp(argv[i], "--tune"))
       tune = true;
+    else if (!std::strcmp(argv[i], "--all-gpus"))
+      all_gpus = true;
     else {
       std::printf("mi-stream [--mib N] [--iters N] [--device D] [--no-mfma]"
-                  " [--tune]\n");
+                  " [--tune] [--all-gpus]\n");
       return !std::strcmp(argv[i], "--help") ? 0 : 2;
     }
   }
@@ -70,6 +74,59 @@ int main(int argc, char** argv) {
     std::fprintf(stderr, "mi-stream: no GPUs visible\n");
     return 1;
   }
+
+  if (all_gpus) {
+    // concurrent non-temporal triad on every visible GPU (the in-pod
+    // demonstration of the headline metric at amd.com/gpu: N): per-GPU
+    // buffers + stream, launches overlapped, one global wall-clock.
+    const int64_t n_ = mib * (1 << 20) / 4;
+    const int64_t n4_ = n_ / 4;
+    const double step_bytes = 3.0 * n_ * 4;
+    std::vector<f4*> A(ndev), B(ndev), C(ndev);
+    std::vector<hipStream_t> streams(ndev);
+    for (int d = 0; d < ndev; ++d) {
+      HIP_CHECK(hipSetDevice(d));
+      HIP_CHECK(hipMalloc(&A[d], n_ * 4));
+      HIP_CHECK(hipMalloc(&B[d], n_ * 4));
+      HIP_CHECK(hipMalloc(&C[d], n_ * 4));
+      HIP_CHECK(hipMemset(B[d], 0x3c, n_ * 4));
+      HIP_CHECK(hipMemset(C[d], 0x3d, n_ * 4));
+      HIP_CHECK(hipStreamCreate(&streams[d]));
+    }
+    dim3 g((uint32_t)k3samd_kern::stream_grid(n4_));
+    dim3 blk(k3samd_kern::kThreadsPerBlock);
+    auto launch_all = [&] {
+      for (int d = 0; d < ndev; ++d) {
+        (void)hipSetDevice(d);
+        hipLaunchKernelGGL(k3samd_kern::stream_triad_kernel<true>, g, blk, 0,
+                           streams[d], A[d], B[d], C[d], 2.5f, n4_);
+      }
+    };
+    auto sync_all = [&] {
+      for (int d = 0; d < ndev; ++d) (void)hipStreamSynchronize(streams[d]);
+    };
+    for (int w = 0; w < 3; ++w) launch_all();
+    sync_all();
+    auto t0 = std::chrono::steady_clock::now();
+    for (int it = 0; it < iters; ++it) launch_all();
+    sync_all();
+    double sec = std::chrono::duration<double>(
+                     std::chrono::steady_clock::now() - t0).count();
+    double agg = ndev * step_bytes * iters / sec / 1e9;
+    std::printf("mi-stream all-gpus: %d GPU(s) x %lld MiB, aggregate triad "
+                "%.1f GB/s (%.1f per GPU)\n",
+                ndev, (long long)mib, agg, agg / ndev);
+    std::printf("{\"payload\": \"mi-stream\", \"mode\": \"all-gpus\", "
+                "\"n_gpus\": %d, \"aggregate_triad_gbps\": %.1f}\n",
+                ndev, agg);
+    for (int d = 0; d < ndev; ++d) {
+      (void)hipSetDevice(d);
+      (void)hipFree(A[d]); (void)hipFree(B[d]); (void)hipFree(C[d]);
+      (void)hipStreamDestroy(streams[d]);
+    }
+    return 0;
+  }
+
   HIP_CHECK(hipSetDevice(device));
   hipDeviceProp_t prop;
   HIP_CHECK(hipGetDeviceProperties(&prop, device));
