@@ -34,6 +34,19 @@ def test_mi_stream_runs():
 
 
 @requires_gpu
+def test_mi_stream_all_gpus():
+    proc = subprocess.run([str(BIN / "mi-stream"), "--all-gpus",
+                           "--mib", "256", "--iters", "5"],
+                          capture_output=True, text=True, timeout=300)
+    assert proc.returncode == 0, proc.stderr
+    last = proc.stdout.strip().splitlines()[-1]
+    j = json.loads(last)
+    assert j["mode"] == "all-gpus"
+    assert j["n_gpus"] >= 1
+    assert j["aggregate_triad_gbps"] > 4000 * j["n_gpus"]
+
+
+@requires_gpu
 def test_mi_allreduce_single_gpu():
     proc = subprocess.run([str(BIN / "mi-allreduce"), "--ngpus", "1",
                            "--min-mib", "4", "--max-mib", "16",
