@@ -249,6 +249,30 @@ def test_rename_by_default(harness):
     assert h.kubelet.requests[0]["resource_name"] == "amd.com/gpu.shared"
 
 
+def test_allocate_multi_container_pod(harness):
+    """A pod with two GPU containers sends one AllocateRequest with two
+    container_requests; each gets its own response."""
+    h = harness(n_gpus=4, replicas=1, register=False)
+    devs = pb.decode_list_and_watch(next(h.stream("ListAndWatch")))
+    req = pb.encode_allocate_request([[devs[0]["id"]], [devs[1]["id"]]])
+    resp = pb.decode_allocate_response(h.call("Allocate", req))
+    assert len(resp) == 2
+    assert resp[0]["envs"]["K3SAMD_RENDER_MINORS"] == "128"
+    assert resp[1]["envs"]["K3SAMD_RENDER_MINORS"] == "129"
+
+
+def test_preferred_allocation_must_include(harness):
+    h = harness(n_gpus=2, replicas=4, register=False)
+    devs = pb.decode_list_and_watch(next(h.stream("ListAndWatch")))
+    avail = [d["id"] for d in devs]
+    must = [avail[5]]  # a replica of GPU 1
+    resp = pb.decode_preferred_response(
+        h.call("GetPreferredAllocation",
+               pb.encode_preferred_request(avail, must, 2)))
+    assert len(resp[0]) == 2
+    assert must[0] in resp[0]
+
+
 def test_preferred_allocation_packs_replicas(harness):
     h = harness(n_gpus=2, replicas=4, register=False)
     devs = pb.decode_list_and_watch(next(h.stream("ListAndWatch")))
