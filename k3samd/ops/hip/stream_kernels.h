@@ -167,7 +167,11 @@ static __global__ void mfma_throughput32_kernel(float* __restrict__ out,
     b[j] = (short)(0x3f00 + ((threadIdx.x * 3 + j) & 7));
   }
   f32x16 acc0 = {}, acc1 = {};
-  for (int i = 0; i < iters; ++i) {
+  // 2 chains x 2-deep unroll: hides loop scalar overhead behind the
+  // 64-cycle MFMA issue slots
+  for (int i = 0; i < iters; i += 2) {
+    acc0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc0, 0, 0, 0);
+    acc1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc1, 0, 0, 0);
     acc0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc0, 0, 0, 0);
     acc1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc1, 0, 0, 0);
   }
