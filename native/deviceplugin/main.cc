@@ -80,6 +80,7 @@ int main(int argc, char** argv) {
 
   cfg.use_cdi = use_cdi;
   k3samd::DevicePlugin plugin(cfg, k3samd::default_sysfs_root());
+  if (!config_path.empty()) plugin.watch_config(config_path);
 
   if (oneshot) {
     k3samd::JsonWriter w;

@@ -3,7 +3,9 @@
 #include <algorithm>
 #include <chrono>
 #include <cstdio>
+#include <fstream>
 #include <set>
+#include <sstream>
 #include <sys/stat.h>
 #include <thread>
 #include <tuple>
@@ -70,7 +72,7 @@ DevicePlugin::~DevicePlugin() { stop(); }
 
 void DevicePlugin::refresh_topology() {
   Topology topo = enumerate_topology(sysfs_root_);
-  std::lock_guard<std::mutex> lk(mu_);
+  std::lock_guard<std::recursive_mutex> lk(mu_);
   topo_ = std::move(topo);
   devices_.clear();
   for (size_t gi = 0; gi < topo_.gpus.size(); ++gi) {
@@ -88,14 +90,16 @@ void DevicePlugin::refresh_topology() {
 }
 
 std::vector<VirtualDevice> DevicePlugin::devices() const {
-  std::lock_guard<std::mutex> lk(mu_);
+  std::lock_guard<std::recursive_mutex> lk(mu_);
   return devices_;
 }
 
 std::string DevicePlugin::advertised_resource() const {
   // NVIDIA-plugin semantics (README.md:112 behavior surface): with sharing
   // enabled and renameByDefault, the resource is advertised as
-  // "<name>.shared".
+  // "<name>.shared". cfg_ can be rewritten by config hot-reload, so read
+  // under the lock (recursive: callers may already hold it).
+  std::lock_guard<std::recursive_mutex> lk(mu_);
   if (cfg_.replicas > 1 && cfg_.rename_by_default)
     return cfg_.resource_name + ".shared";
   return cfg_.resource_name;
@@ -109,7 +113,7 @@ std::string DevicePlugin::handle_options() {
 }
 
 std::vector<dp::Device> DevicePlugin::current_device_list() {
-  std::lock_guard<std::mutex> lk(mu_);
+  std::lock_guard<std::recursive_mutex> lk(mu_);
   std::vector<dp::Device> out;
   for (const auto& vd : devices_) {
     dp::Device d;
@@ -133,7 +137,7 @@ GrpcStatus DevicePlugin::handle_allocate(const std::string& req,
   if (!dp::decode_allocate_request(req, containers))
     return {13, "malformed AllocateRequest"};
 
-  std::lock_guard<std::mutex> lk(mu_);
+  std::lock_guard<std::recursive_mutex> lk(mu_);
   std::vector<dp::ContainerAllocateResponse> crs;
   for (const auto& ids : containers) {
     if (cfg_.replicas > 1 && cfg_.fail_requests_greater_than_one &&
@@ -203,7 +207,7 @@ GrpcStatus DevicePlugin::handle_preferred(const std::string& req,
   if (!dp::decode_preferred_request(req, reqs))
     return {13, "malformed PreferredAllocationRequest"};
 
-  std::lock_guard<std::mutex> lk(mu_);
+  std::lock_guard<std::recursive_mutex> lk(mu_);
   std::vector<std::vector<std::string>> out;
   for (const auto& pr : reqs) {
     std::vector<std::string> chosen(pr.must_include);
@@ -231,12 +235,59 @@ GrpcStatus DevicePlugin::handle_preferred(const std::string& req,
   return GrpcStatus::Ok();
 }
 
+void DevicePlugin::watch_config(const std::string& path) {
+  config_path_ = path;
+  struct stat st{};
+  if (::stat(path.c_str(), &st) == 0) {
+    config_mtime_ = st.st_mtim.tv_sec * 1000000000L + st.st_mtim.tv_nsec;
+    config_size_ = (long)st.st_size;
+  }
+}
+
+bool DevicePlugin::poll_config_once() {
+  if (config_path_.empty()) return false;
+  struct stat st{};
+  if (::stat(config_path_.c_str(), &st) != 0) return false;
+  long mtime_ns = st.st_mtim.tv_sec * 1000000000L + st.st_mtim.tv_nsec;
+  if (mtime_ns == config_mtime_ && (long)st.st_size == config_size_)
+    return false;
+  config_mtime_ = mtime_ns;
+  config_size_ = (long)st.st_size;
+
+  std::ifstream f(config_path_);
+  std::stringstream ss;
+  ss << f.rdbuf();
+  PluginConfig fresh;
+  fresh.resource_name = cfg_.resource_name;
+  fresh.use_cdi = cfg_.use_cdi;
+  fresh.cdi_kind = cfg_.cdi_kind;
+  std::string err;
+  if (!PluginConfig::from_yaml(ss.str(), fresh, &err)) {
+    std::fprintf(stderr, "deviceplugin: ignoring invalid config update: %s\n",
+                 err.c_str());
+    return false;
+  }
+  std::string old_resource = advertised_resource();
+  {
+    std::lock_guard<std::recursive_mutex> lk(mu_);
+    cfg_ = fresh;
+  }
+  refresh_topology();  // rebuilds devices with the new replicas; bumps gen
+  std::fprintf(stderr,
+               "deviceplugin: config reloaded (replicas=%d, %zu devices)\n",
+               fresh.replicas, devices().size());
+  // a renamed resource needs a fresh registration with kubelet
+  if (!kubelet_sock_.empty() && advertised_resource() != old_resource)
+    register_with_kubelet();
+  return true;
+}
+
 bool DevicePlugin::poll_health_once() {
   Topology topo = enumerate_topology(sysfs_root_);
   std::set<std::string> present;
   for (const auto& g : topo.gpus) present.insert(g.stable_id());
 
-  std::lock_guard<std::mutex> lk(mu_);
+  std::lock_guard<std::recursive_mutex> lk(mu_);
   bool changed = false;
   for (auto& vd : devices_) {
     const auto& g = topo_.gpus[vd.gpu_index];
@@ -257,7 +308,7 @@ GrpcStatus DevicePlugin::handle_list_and_watch(
     const std::string&, const GrpcServer::WriteFn& write) {
   uint64_t seen_gen;
   {
-    std::lock_guard<std::mutex> lk(mu_);
+    std::lock_guard<std::recursive_mutex> lk(mu_);
     seen_gen = generation_;
   }
   metrics_.list_and_watch_updates_total.fetch_add(1);
@@ -270,7 +321,7 @@ GrpcStatus DevicePlugin::handle_list_and_watch(
     std::this_thread::sleep_for(std::chrono::milliseconds(100));
     uint64_t gen;
     {
-      std::lock_guard<std::mutex> lk(mu_);
+      std::lock_guard<std::recursive_mutex> lk(mu_);
       gen = generation_;
     }
     if (gen == seen_gen) continue;
@@ -284,7 +335,7 @@ GrpcStatus DevicePlugin::handle_list_and_watch(
 std::string DevicePlugin::render_metrics() {
   size_t healthy = 0, unhealthy = 0;
   {
-    std::lock_guard<std::mutex> lk(mu_);
+    std::lock_guard<std::recursive_mutex> lk(mu_);
     for (const auto& vd : devices_) (vd.healthy ? healthy : unhealthy)++;
   }
   char buf[2048];
@@ -397,6 +448,7 @@ bool DevicePlugin::serve(const std::string& plugin_sock,
       while (!stopping_.load()) {
         std::this_thread::sleep_for(std::chrono::milliseconds(health_poll_ms));
         if (stopping_.load()) break;
+        poll_config_once();
         poll_health_once();
       }
     });

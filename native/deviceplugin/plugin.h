@@ -53,6 +53,13 @@ class DevicePlugin {
   // Re-enumerate the topology and rebuild the advertised device list.
   void refresh_topology();
 
+  // Watch `path` for config changes (mtime/size) from the health loop;
+  // a valid new config rebuilds the device list (pushed over
+  // ListAndWatch); an invalid one is logged and ignored.
+  void watch_config(const std::string& path);
+  // exposed for tests: returns true if the config was reloaded
+  bool poll_config_once();
+
   std::vector<VirtualDevice> devices() const;
   // resource name after renameByDefault is applied
   std::string advertised_resource() const;
@@ -84,8 +91,11 @@ class DevicePlugin {
   PluginConfig cfg_;
   std::string sysfs_root_;
   std::string dev_root_;
+  std::string config_path_;
+  long config_mtime_ = -1;
+  long config_size_ = -1;
 
-  mutable std::mutex mu_;
+  mutable std::recursive_mutex mu_;
   Topology topo_;
   std::vector<VirtualDevice> devices_;
   uint64_t generation_ = 0;  // bumped on device-list changes

@@ -298,6 +298,28 @@ def test_health_transition_pushes_update(harness):
     assert by_health == ["Healthy", "Unhealthy"]
 
 
+def test_config_hot_reload(harness, tmp_path):
+    """Editing the mounted config (kubelet updates ConfigMap mounts in
+    place) re-fans-out the device list and pushes it over the live
+    ListAndWatch stream — allocatable changes without a pod restart."""
+    h = harness(n_gpus=2, replicas=1, register=False, health_poll_ms=100)
+    stream = h.stream("ListAndWatch", timeout=30)
+    first = pb.decode_list_and_watch(next(stream))
+    assert len(first) == 2
+    cfg = tmp_path / "config.yaml"
+    time.sleep(0.05)
+    cfg.write_text(DEFAULT_CFG.format(replicas=4))
+    second = pb.decode_list_and_watch(next(stream))
+    assert len(second) == 8
+    assert all("::" in d["id"] for d in second)
+    # invalid update is ignored, plugin stays up with the last-good config
+    time.sleep(0.05)
+    cfg.write_text("version: v99\n")
+    time.sleep(0.5)
+    devs = pb.decode_list_and_watch(next(h.stream("ListAndWatch")))
+    assert len(devs) == 8
+
+
 def test_reregister_after_kubelet_restart(harness):
     """kubelet restarts drop plugin registrations; the plugin must notice
     the new registration socket and Register again."""
