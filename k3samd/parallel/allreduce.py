@@ -76,8 +76,8 @@ def busbw_sweep(sizes_bytes, iters=20, device=None, group=None,
         per_op = float(e[0]) / iters
         algbw = n * 4 / per_op / 1e9
         busbw = algbw * factor
-        rows.append({"bytes": n * 4, "algbw_gbps": round(algbw, 2),
-                     "busbw_gbps": round(busbw, 2)})
+        rows.append({"bytes": n * 4, "algbw_gbps": round(algbw, 4),
+                     "busbw_gbps": round(busbw, 4)})
     return rows
 
 

@@ -1,6 +1,9 @@
 #include "grpc_transport.h"
 
+#include <cerrno>
 #include <chrono>
+#include <csignal>
+#include <cstdio>
 #include <cstring>
 #include <condition_variable>
 #include <poll.h>
@@ -68,7 +71,9 @@ bool read_full(int fd, void* buf, size_t n, int timeout_ms = -1) {
 bool write_full(int fd, const void* buf, size_t n) {
   const uint8_t* p = (const uint8_t*)buf;
   while (n > 0) {
-    ssize_t r = ::write(fd, p, n);
+    // MSG_NOSIGNAL: a peer that closed mid-write must surface as EPIPE,
+    // not a process-killing SIGPIPE (churned gRPC channels do this)
+    ssize_t r = ::send(fd, p, n, MSG_NOSIGNAL);
     if (r <= 0) return false;
     p += r;
     n -= (size_t)r;
@@ -227,6 +232,8 @@ void GrpcServer::add_server_stream(const std::string& path, StreamHandler h) {
 }
 
 bool GrpcServer::start(const std::string& unix_path) {
+  // belt-and-braces with MSG_NOSIGNAL: never die on peer-closed sockets
+  std::signal(SIGPIPE, SIG_IGN);
   ::unlink(unix_path.c_str());
   int fd = ::socket(AF_UNIX, SOCK_STREAM | SOCK_CLOEXEC, 0);
   if (fd < 0) return false;
@@ -342,7 +349,13 @@ void GrpcServer::serve_loop() {
     int lfd = listen_fd_.load();
     if (lfd < 0) break;
     int cfd = ::accept(lfd, nullptr, nullptr);
-    if (cfd < 0) break;
+    if (cfd < 0) {
+      if (errno == EINTR || errno == ECONNABORTED) continue;
+      if (!stopping_.load())
+        std::fprintf(stderr, "grpc-server: accept failed: %s\n",
+                     std::strerror(errno));
+      break;
+    }
     auto conn = std::make_shared<H2Conn>();
     conn->fd = cfd;
     {
@@ -354,6 +367,9 @@ void GrpcServer::serve_loop() {
       char preface[kPrefaceLen];
       if (!read_full(conn->fd, preface, kPrefaceLen, 10000) ||
           std::memcmp(preface, kPreface, kPrefaceLen) != 0) {
+        if (std::getenv("K3SAMD_GRPC_DEBUG"))
+          std::fprintf(stderr, "grpc-server: conn fd=%d preface fail errno=%s\n",
+                       conn->fd, std::strerror(errno));
         conn->close_fd();
       } else {
         conn->send(frame_bytes(kSettings, 0, 0, {}));  // our (default) settings

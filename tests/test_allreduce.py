@@ -40,8 +40,9 @@ for r in rows:
 
 # other collectives run and produce sane numbers on gloo too
 for op in ("all_gather", "broadcast"):
-    rows = busbw_sweep([1 << 14], iters=2, op=op)
-    assert rows[0]["busbw_gbps"] > 0, op
+    rows = busbw_sweep([1 << 18], iters=2, op=op)
+    assert rows[0]["bytes"] == 1 << 18
+    assert rows[0]["busbw_gbps"] > 0, (op, rows)
 if rank == 0:
     print("WORKER_OK", json.dumps(rows))
 dist.destroy_process_group()
