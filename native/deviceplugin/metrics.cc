@@ -83,8 +83,8 @@ void MetricsServer::serve_loop() {
                           "Content-Type: text/plain; version=0.0.4\r\n"
                           "Content-Length: %zu\r\n\r\n",
                           body.size());
-    (void)::write(cfd, head, (size_t)n);
-    (void)::write(cfd, body.data(), body.size());
+    (void)::send(cfd, head, (size_t)n, MSG_NOSIGNAL);
+    (void)::send(cfd, body.data(), body.size(), MSG_NOSIGNAL);
     ::close(cfd);
   }
 }
