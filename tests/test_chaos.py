@@ -1,0 +1,121 @@
+"""Chaos tests: hostile/broken byte streams against the gRPC server must
+never take the device plugin down (the SIGPIPE churn crash was exactly
+this class of bug)."""
+
+import socket
+import struct
+import subprocess
+import time
+from pathlib import Path
+
+import grpc
+import pytest
+
+import pb_v1beta1 as pb
+from sysfs_builder import build_tree
+
+REPO = Path(__file__).resolve().parent.parent
+PLUGIN = REPO / "native" / "bin" / "k3samd-device-plugin"
+PREFACE = b"PRI * HTTP/2.0\r\n\r\nSM\r\n\r\n"
+
+IDENT = lambda b: b  # noqa: E731
+
+
+@pytest.fixture(scope="session", autouse=True)
+def build_native():
+    subprocess.run(["make", "-C", str(REPO / "native"), "-j8"], check=True,
+                   capture_output=True)
+
+
+@pytest.fixture
+def plugin(tmp_path):
+    root = build_tree(tmp_path / "sys", n_gpus=1)
+    sock = str(tmp_path / "amd.sock")
+    proc = subprocess.Popen(
+        [str(PLUGIN), "--plugin-sock", sock, "--no-register",
+         "--health-poll-ms", "0"],
+        env={"K3SAMD_SYSFS_ROOT": str(root)},
+        stdout=subprocess.PIPE, stderr=subprocess.PIPE)
+    deadline = time.time() + 10
+    while not Path(sock).exists():
+        assert time.time() < deadline and proc.poll() is None
+        time.sleep(0.05)
+    yield sock, proc
+    proc.terminate()
+    proc.wait(timeout=10)
+
+
+def raw(sock_path, payload, linger=0.05):
+    s = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+    s.connect(sock_path)
+    if payload:
+        s.sendall(payload)
+    time.sleep(linger)
+    s.close()
+
+
+def frame(ftype, flags, stream, payload):
+    return struct.pack(">I", len(payload))[1:] + bytes([ftype, flags]) + \
+        struct.pack(">I", stream) + payload
+
+
+def assert_alive(sock_path, proc):
+    assert proc.poll() is None, proc.stderr.read()[-500:]
+    ch = grpc.insecure_channel(f"unix:{sock_path}")
+    opt = ch.unary_unary("/v1beta1.DevicePlugin/GetDevicePluginOptions",
+                         request_serializer=IDENT,
+                         response_deserializer=IDENT)
+    opts = pb.decode_options(opt(b"", timeout=10))
+    assert opts["get_preferred_allocation_available"] is True
+    ch.close()
+
+
+def test_survives_hostile_streams(plugin):
+    sock, proc = plugin
+    cases = [
+        b"",                                   # connect + close
+        b"\x00" * 64,                          # garbage, no preface
+        b"GET / HTTP/1.1\r\n\r\n",             # wrong protocol
+        PREFACE,                               # preface then hangup
+        PREFACE + b"\xff" * 32,                # bogus frames
+        PREFACE + frame(0x1, 0x4, 1, b"\xff\xfe\xfd"),   # invalid HPACK
+        PREFACE + frame(0x1, 0x4, 1, b"")[:7],           # truncated header
+        PREFACE + b"\xff\xff\xff\x00\x00\x00\x00\x00\x01",  # 16MB len claim
+        PREFACE + frame(0x6, 0x0, 0, b"12345678") * 50,  # ping flood
+        PREFACE + frame(0x9, 0x4, 1, b"x"),    # CONTINUATION w/o HEADERS
+    ]
+    for payload in cases:
+        for _ in range(3):
+            raw(sock, payload)
+    assert_alive(sock, proc)
+
+
+def test_survives_disconnect_during_response(plugin):
+    """Peer disappears right after sending a valid request: the server's
+    response writes hit a closed socket (the EPIPE path)."""
+    sock, proc = plugin
+    from test_http2_raw import request_headers, grpc_frame
+    for _ in range(10):
+        s = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+        s.connect(sock)
+        s.sendall(PREFACE + frame(0x4, 0, 0, b"") +
+                  frame(0x1, 0x4, 1,
+                        request_headers(
+                            b"/v1beta1.DevicePlugin/GetDevicePluginOptions")) +
+                  frame(0x0, 0x1, 1, grpc_frame(b"")))
+        s.close()  # gone before reading the response
+    time.sleep(0.3)
+    assert_alive(sock, proc)
+
+
+def test_survives_many_half_open(plugin):
+    sock, proc = plugin
+    socks = []
+    for _ in range(30):
+        s = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+        s.connect(sock)
+        socks.append(s)
+    time.sleep(0.2)
+    for s in socks:
+        s.close()
+    assert_alive(sock, proc)
