@@ -532,8 +532,20 @@ void GrpcServer::serve_loop() {
               }
             }
           }
+          auto finish_stream = [conn](const std::shared_ptr<H2Stream>& st) {
+            // a finished RPC's state is dropped from the map (kubelet keeps
+            // one connection for the plugin's lifetime; per-RPC streams
+            // must not accumulate)
+            std::lock_guard<std::mutex> lk(conn->mu);
+            conn->streams.erase(st->id);
+          };
           for (auto& st : ready) {
             std::string path = find_header(st->headers, ":path");
+            if (st->cancelled.load()) {  // e.g. over-size request
+              send_trailers(conn, st, {8, "request too large"}, true);
+              finish_stream(st);
+              continue;
+            }
             std::string req;
             first_grpc_message(st->data, req);
             auto uit = unary_.find(path);
@@ -547,12 +559,13 @@ void GrpcServer::serve_loop() {
               } else {
                 send_trailers(conn, st, gs, true);
               }
+              finish_stream(st);
               continue;
             }
             auto sit = stream_.find(path);
             if (sit != stream_.end()) {
               StreamHandler handler = sit->second;
-              spawn([conn, st, handler, req] {
+              spawn([conn, st, handler, finish_stream, req] {
                 send_response_headers(conn, st);
                 auto write = [conn, st](const std::string& msg) -> bool {
                   if (conn->closed.load() || st->cancelled.load()) return false;
@@ -561,10 +574,12 @@ void GrpcServer::serve_loop() {
                 GrpcStatus gs = handler(req, write);
                 if (!conn->closed.load() && !st->cancelled.load())
                   send_trailers(conn, st, gs, false);
+                finish_stream(st);
               });
               continue;
             }
             send_trailers(conn, st, {12, "unknown method " + path}, true);
+            finish_stream(st);
           }
         }
       }
