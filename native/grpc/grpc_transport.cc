@@ -134,7 +134,6 @@ struct H2Stream {
   std::vector<Header> headers;
   std::string data;
   bool request_complete = false;
-  bool headers_done = false;
   std::atomic<bool> cancelled{false};
   std::atomic<int> send_window{kDefaultWindow};
   bool dispatched = false;
@@ -510,7 +509,6 @@ void GrpcServer::serve_loop() {
             auto st = std::make_shared<H2Stream>();
             st->id = header_stream;
             if (!conn->decoder.decode(header_accum, st->headers)) break;
-            st->headers_done = true;
             st->send_window.store(conn->peer_initial_window);
             if (header_flags & kEndStream) st->request_complete = true;
             {

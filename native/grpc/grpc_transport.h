@@ -55,7 +55,6 @@ class GrpcServer {
   bool running() const { return running_.load(); }
 
  private:
-  struct Impl;
   void serve_loop();
 
   std::map<std::string, UnaryHandler> unary_;

@@ -79,7 +79,6 @@ class PluginHarness:
     def __init__(self, tmp_path, n_gpus=8, replicas=1, cfg_text=None,
                  register=True, health_poll_ms=200):
         self.root = build_tree(tmp_path / "sys", n_gpus=n_gpus)
-        self.dir = tmp_path
         cfg = tmp_path / "config.yaml"
         cfg.write_text(cfg_text or DEFAULT_CFG.format(replicas=replicas))
         self.plugin_sock = str(tmp_path / "amd-gpu.sock")
