@@ -9,8 +9,9 @@ capabilities of the K3S-NVidia reference stack (/root/reference):
   MFMA matrix-core smoke) used by the in-pod GPU payloads and bench.py.
 * ``k3samd.parallel`` — RCCL-over-xGMI collective smoke paths
   (torch.distributed; gloo on CPU for tests).
-* ``k3samd.utils`` — pure-python KFD sysfs topology parsing (test mirror
-  of the C++ library).
+* ``k3samd.utils`` — profiling tooling (rocprofv3 rocpd database
+  summarizer); topology truth lives in the C++ library (native/topology),
+  exposed to Python via the binaries' ``--json`` outputs.
 * ``deploy/`` — Helm chart + manifests with the same values surface as the
   reference (values.yaml:1-18).
 """
