@@ -43,7 +43,8 @@ struct CardStats {
   uint64_t vram_used = 0;
   uint64_t vram_total = 0;
   long busy_percent = -1;
-  long temp_mc = -1;  // millidegrees C
+  long temp_mc = -1;   // millidegrees C
+  long power_uw = -1;  // microwatts
 };
 
 CardStats card_stats(const std::string& sysfs_root, int card_index) {
@@ -60,10 +61,15 @@ CardStats card_stats(const std::string& sysfs_root, int card_index) {
   if (DIR* d = ::opendir(hwdir.c_str())) {
     while (struct dirent* e = ::readdir(d)) {
       if (e->d_name[0] == '.') continue;
-      std::string t = read_trim(hwdir + "/" + e->d_name + "/temp1_input");
-      if (!t.empty()) {
+      std::string base = hwdir + "/" + e->d_name;
+      std::string t = read_trim(base + "/temp1_input");
+      if (!t.empty() && st.temp_mc < 0)
         st.temp_mc = std::strtol(t.c_str(), nullptr, 10);
-        break;
+      // amdgpu exposes average or instantaneous package power
+      for (const char* f : {"/power1_average", "/power1_input"}) {
+        std::string pwr = read_trim(base + f);
+        if (!pwr.empty() && st.power_uw < 0)
+          st.power_uw = std::strtol(pwr.c_str(), nullptr, 10);
       }
     }
     ::closedir(d);
@@ -111,6 +117,7 @@ int main(int argc, char** argv) {
       w.key("numa_node").value(g.numa_node);
       w.key("busy_percent").value((int64_t)st.busy_percent);
       w.key("temp_milli_c").value((int64_t)st.temp_mc);
+      w.key("power_uw").value((int64_t)st.power_uw);
       w.end_obj();
     }
     w.end_arr();
@@ -122,9 +129,9 @@ int main(int argc, char** argv) {
   std::printf("+---------------------------------------------------------------------------+\n");
   std::printf("| mi355x-smi            driver: %-12s                 k3samd stack   |\n",
               topo.driver_version.empty() ? "unknown" : topo.driver_version.c_str());
-  std::printf("+----+----------------------+--------+-----+-------------------+------+-----+\n");
-  std::printf("| ## | Name                 | Arch   | CUs | VRAM used / total | xGMI | Tmp |\n");
-  std::printf("+----+----------------------+--------+-----+-------------------+------+-----+\n");
+  std::printf("+----+----------------------+--------+-----+-------------------+------+-----+------+\n");
+  std::printf("| ## | Name                 | Arch   | CUs | VRAM used / total | xGMI | Tmp | Pwr  |\n");
+  std::printf("+----+----------------------+--------+-----+-------------------+------+-----+------+\n");
   int i = 0;
   for (const auto& g : topo.gpus) {
     CardStats st = card_stats(root, g.card_index);
@@ -136,14 +143,19 @@ int main(int argc, char** argv) {
       std::snprintf(temp, sizeof(temp), "%2ldC", st.temp_mc / 1000);
     else
       std::snprintf(temp, sizeof(temp), " - ");
-    std::printf("| %2d | %-20.20s | %-6s | %3u | %-17s | %4d | %s |\n", i++,
-                g.name.c_str(), g.gfx_arch().c_str(), g.compute_units(), vram,
-                g.xgmi_links, temp);
+    char pwr[16];
+    if (st.power_uw >= 0)
+      std::snprintf(pwr, sizeof(pwr), "%4ldW", st.power_uw / 1000000);
+    else
+      std::snprintf(pwr, sizeof(pwr), "  - ");
+    std::printf("| %2d | %-20.20s | %-6s | %3u | %-17s | %4d | %s | %s |\n",
+                i++, g.name.c_str(), g.gfx_arch().c_str(), g.compute_units(),
+                vram, g.xgmi_links, temp, pwr);
   }
   if (topo.gpus.empty()) {
     std::printf("| no AMD GPUs found (no KFD topology under %s)\n",
                 root.c_str());
   }
-  std::printf("+----+----------------------+--------+-----+-------------------+------+-----+\n");
+  std::printf("+----+----------------------+--------+-----+-------------------+------+-----+------+\n");
   return 0;
 }

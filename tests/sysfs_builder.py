@@ -116,5 +116,6 @@ def build_tree(
             hw = card / "hwmon/hwmon0"
             hw.mkdir(parents=True, exist_ok=True)
             (hw / "temp1_input").write_text("53000\n")
+            (hw / "power1_average").write_text("135000000\n")  # 135 W
 
     return root

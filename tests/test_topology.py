@@ -50,6 +50,7 @@ def test_eight_gpu_node(native_bin, tmp_path):
         assert g["id"].startswith("amdgpu-1a2b3c4d5e6f")
         assert g["busy_percent"] == 0
         assert g["temp_milli_c"] == 53000
+        assert g["power_uw"] == 135000000
 
 
 def test_cpu_only_node(native_bin, tmp_path):
