@@ -562,6 +562,19 @@ void GrpcServer::serve_loop() {
             }
             auto sit = stream_.find(path);
             if (sit != stream_.end()) {
+              // cap concurrent server-streams per connection (each owns a
+              // thread; kubelet uses exactly one ListAndWatch)
+              size_t live;
+              {
+                std::lock_guard<std::mutex> lk(conn->mu);
+                live = conn->streams.size();
+              }
+              if (live > 64) {
+                send_trailers(conn, st, {8, "too many concurrent streams"},
+                              true);
+                finish_stream(st);
+                continue;
+              }
               StreamHandler handler = sit->second;
               spawn([conn, st, handler, finish_stream, req] {
                 send_response_headers(conn, st);
