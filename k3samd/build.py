@@ -66,11 +66,9 @@ def build_extension(verbose: bool = True) -> Path:
     # torch's load() with is_python_module=False loads the .so; we want the
     # artifact in-tree under the package so imports and the gpurun snapshot
     # find it.  Find the built library and copy it.
-    built = None
-    for cand in build_dir.glob("*.so"):
-        built = cand
-    if built is None:
-        raise RuntimeError("extension build produced no .so")
+    built = build_dir / f"{EXT_NAME}.so"
+    if not built.exists():
+        raise RuntimeError(f"extension build produced no {built}")
     import shutil
 
     shutil.copy2(built, so)

@@ -147,6 +147,28 @@ def test_rocm_mount_injection(tmp_path):
     assert "ro" in rocm_mounts[0]["options"]
 
 
+def test_rocm_default_with_env_optout(tmp_path):
+    '''K3SAMD_INJECT_ROCM=0 overrides an inject-by-default runtime config.'''
+    rocm = tmp_path / "opt-rocm"
+    rocm.mkdir()
+    sysfs = build_tree(tmp_path / "sys", n_gpus=1)
+    cfg = tmp_path / "config.json"
+    spec = json.loads(json.dumps(BASE_SPEC))
+    spec["process"]["env"].append("K3SAMD_INJECT_ROCM=0")
+    cfg.write_text(json.dumps(spec))
+    subprocess.run(
+        [str(RUNTIME), "--transform-only", str(cfg)],
+        env={"K3SAMD_SYSFS_ROOT": str(sysfs),
+             "K3SAMD_DEV_ROOT": str(tmp_path / "nodev"),
+             "K3SAMD_ROCM_ROOT": str(rocm),
+             "K3SAMD_INJECT_ROCM_DEFAULT": "1"},
+        check=True, capture_output=True, timeout=60)
+    out = json.loads(cfg.read_text())
+    assert not any(m["destination"] == "/opt/rocm" for m in out["mounts"])
+    # devices still injected
+    assert "/dev/kfd" in device_paths(out)
+
+
 def test_cpu_only_node_injects_nothing(tmp_path):
     out, _ = transform(tmp_path, BASE_SPEC, n_gpus=0)
     assert "/dev/kfd" not in json.dumps(out)
