@@ -319,6 +319,20 @@ def test_config_hot_reload(harness, tmp_path):
     assert len(devs) == 8
 
 
+def test_hot_reload_rename_reregisters(harness, tmp_path):
+    '''Turning renameByDefault on via config reload changes the advertised
+    resource, which requires a fresh Register with kubelet.'''
+    h = harness(n_gpus=1, replicas=4, register=True, health_poll_ms=100)
+    assert h.kubelet.event.wait(5)
+    assert h.kubelet.requests[0]["resource_name"] == "amd.com/gpu"
+    h.kubelet.event.clear()
+    cfg = tmp_path / "config.yaml"
+    cfg.write_text(DEFAULT_CFG.format(replicas=4).replace(
+        "renameByDefault: false", "renameByDefault: true"))
+    assert h.kubelet.event.wait(10), "no re-register after rename"
+    assert h.kubelet.requests[-1]["resource_name"] == "amd.com/gpu.shared"
+
+
 def test_reregister_after_kubelet_restart(harness):
     """kubelet restarts drop plugin registrations; the plugin must notice
     the new registration socket and Register again."""
