@@ -128,6 +128,17 @@ bool oci_inject_gpus(const JPtr& config, const Topology& topo,
   std::string minors_env = env_lookup(env, "K3SAMD_RENDER_MINORS");
   std::string inject_rocm_env = env_lookup(env, "K3SAMD_INJECT_ROCM");
 
+  // fallback: the device plugin also stamps its allocation as an
+  // annotation (k3samd.ai/allocated-gpus); some runtimes propagate
+  // annotations but scrub env
+  if (visible.empty() && minors_env.empty()) {
+    JPtr ann = config->get("annotations");
+    if (ann && ann->type == JValue::kObject) {
+      JPtr a = ann->get("k3samd.ai/allocated-gpus");
+      if (a && a->type == JValue::kString) visible = a->str;
+    }
+  }
+
   if (visible == "none" || visible == "void") {
     if (report) report->skipped = true;
     return true;

@@ -149,6 +149,11 @@ def test_register_handshake(harness):
     assert req["options"]["get_preferred_allocation_available"] is True
 
 
+def test_prestart_container(harness):
+    h = harness(n_gpus=1, register=False)
+    assert h.call("PreStartContainer", b"") == b""
+
+
 def test_options(harness):
     h = harness(n_gpus=2, register=False)
     opts = pb.decode_options(h.call("GetDevicePluginOptions"))

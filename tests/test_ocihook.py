@@ -169,6 +169,16 @@ def test_rocm_default_with_env_optout(tmp_path):
     assert "/dev/kfd" in device_paths(out)
 
 
+def test_annotation_fallback_selection(tmp_path):
+    '''With env scrubbed, the Allocate annotation still scopes injection.'''
+    spec = json.loads(json.dumps(BASE_SPEC))
+    spec["annotations"] = {
+        "k3samd.ai/allocated-gpus": "amdgpu-1a2b3c4d5e6f0001"}
+    out, _ = transform(tmp_path, spec, n_gpus=4)
+    assert device_paths(out) == ["/dev/kfd", "/dev/dri/renderD129",
+                                 "/dev/dri/card1"]
+
+
 def test_cpu_only_node_injects_nothing(tmp_path):
     out, _ = transform(tmp_path, BASE_SPEC, n_gpus=0)
     assert "/dev/kfd" not in json.dumps(out)
