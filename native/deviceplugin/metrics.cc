@@ -75,7 +75,8 @@ void MetricsServer::serve_loop() {
     if (cfd < 0) break;
     // read whatever request line arrives (we answer every request the same)
     char buf[1024];
-    (void)::read(cfd, buf, sizeof(buf));
+    ssize_t ignored = ::read(cfd, buf, sizeof(buf));
+    (void)ignored;
     std::string body = render_ ? render_() : "";
     char head[160];
     int n = std::snprintf(head, sizeof(head),

@@ -9,8 +9,8 @@
 // point for third-party label producers — so the k3samd DaemonSet needs no
 // API-server credentials:
 //
-//   k3samd-node-labeller --features-file \
-//       /etc/kubernetes/node-feature-discovery/features.d/k3samd
+//   k3samd-node-labeller
+//       --features-file=/etc/kubernetes/node-feature-discovery/features.d/k3samd
 //
 // Labels (amd.com domain, mirrors the nvidia.com/gpu.* surface):
 //   amd.com/gpu.present=true        amd.com/gpu.count=8

@@ -138,9 +138,9 @@ int main(int argc, char** argv) {
     char vram[40];
     std::snprintf(vram, sizeof(vram), "%6.0f/%6.0f MiB", mib(st.vram_used),
                   mib(g.vram_bytes));
-    char temp[16];
+    char temp[24];
     if (st.temp_mc >= 0)
-      std::snprintf(temp, sizeof(temp), "%2ldC", st.temp_mc / 1000);
+      std::snprintf(temp, sizeof(temp), "%2ldC", (long)(st.temp_mc / 1000));
     else
       std::snprintf(temp, sizeof(temp), " - ");
     char pwr[16];
