@@ -23,6 +23,7 @@
 #include <sys/sysmacros.h>
 
 #include "../common/json_writer.h"
+#include "../common/version.h"
 #include "../topology/kfd_topology.h"
 
 namespace {
@@ -53,6 +54,7 @@ void emit_device_node(k3samd::JsonWriter& w, const Node& n) {
 }  // namespace
 
 int main(int argc, char** argv) {
+  if (k3samd::handle_version_flag(argc, argv, "k3samd-cdi-gen")) return 0;
   std::string output;
   std::string kind = "amd.com/gpu";
   std::string dev_root = "/dev";

@@ -19,6 +19,7 @@
 #include <sstream>
 
 #include "../common/json_writer.h"
+#include "../common/version.h"
 #include "plugin.h"
 
 namespace {
@@ -36,6 +37,7 @@ std::string read_file(const std::string& p) {
 }  // namespace
 
 int main(int argc, char** argv) {
+  if (k3samd::handle_version_flag(argc, argv, "k3samd-device-plugin")) return 0;
   std::string config_path;
   std::string plugin_sock = "/var/lib/kubelet/device-plugins/amd-gpu.sock";
   std::string kubelet_sock = "/var/lib/kubelet/device-plugins/kubelet.sock";

@@ -31,6 +31,7 @@
 #include <chrono>
 
 #include "../common/json_writer.h"
+#include "../common/version.h"
 #include "../topology/kfd_topology.h"
 
 namespace {
@@ -87,6 +88,7 @@ bool write_features_file(const std::string& path,
 }  // namespace
 
 int main(int argc, char** argv) {
+  if (k3samd::handle_version_flag(argc, argv, "k3samd-node-labeller")) return 0;
   std::string features_file;
   bool oneshot = false, json = false;
   int interval_s = 60;

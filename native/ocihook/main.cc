@@ -78,6 +78,9 @@ bool transform_config(const std::string& path) {
 }  // namespace
 
 int main(int argc, char** argv) {
+  // NOTE: --version is deliberately NOT intercepted — containerd probes
+  // runtimes with `runc --version` and parses runc's output format, so it
+  // must pass through to the real runc below.
   // explicit test mode
   if (argc >= 3 && !std::strcmp(argv[1], "--transform-only")) {
     return transform_config(argv[2]) ? 0 : 1;

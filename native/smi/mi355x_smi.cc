@@ -19,6 +19,7 @@
 #include <string>
 
 #include "../common/json_writer.h"
+#include "../common/version.h"
 #include "../topology/kfd_topology.h"
 
 namespace {
@@ -82,6 +83,7 @@ double mib(uint64_t b) { return (double)b / (1024.0 * 1024.0); }
 }  // namespace
 
 int main(int argc, char** argv) {
+  if (k3samd::handle_version_flag(argc, argv, "mi355x-smi")) return 0;
   bool json = false;
   for (int i = 1; i < argc; ++i) {
     if (!std::strcmp(argv[i], "--json")) json = true;
