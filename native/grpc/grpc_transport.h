@@ -7,8 +7,12 @@
 //   server: unary + server-streaming RPCs (ListAndWatch), flow control,
 //           PING/SETTINGS/WINDOW_UPDATE/RST_STREAM/GOAWAY handling;
 //   client: blocking unary call (Registration.Register).
-// Interoperability is tested against grpcio (tests/test_deviceplugin.py),
-// the same core the real kubelet's grpc-go speaks.
+// Interoperability is tested against grpcio (tests/test_deviceplugin.py,
+// the same gRPC core family kubelet's grpc-go belongs to), against
+// hand-built raw frames for the shapes grpcio never emits (padding,
+// CONTINUATION — tests/test_http2_raw.py), and against hostile byte
+// streams (tests/test_chaos.py). RFC 7541 Appendix C vectors validate the
+// HPACK decoder (native/selftest).
 
 #pragma once
 
