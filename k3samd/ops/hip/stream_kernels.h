@@ -182,6 +182,44 @@ static __global__ void mfma_throughput32_kernel(float* __restrict__ out,
 #endif
 }
 
+#if K3_HAS_MFMA
+using i32x8 = __attribute__((ext_vector_type(8))) int;
+#endif
+
+// Block-scaled MX MFMA throughput (gfx950-only instruction class, the sole
+// path to the chip's low-precision peaks): v_mfma_scale_f32_16x16x128_f8f6f4
+// with FMT selecting the operand format (0 = fp8 e4m3, 4 = fp4). K = 128,
+// so FLOPs per MFMA = 2*16*16*128 = 65536. Scales are E8M0 bias-127 (= 1.0);
+// 4 independent accumulators as in the bf16 kernel.
+template <int FMT>
+static __global__ void mfma_throughput_mx_kernel(float* __restrict__ out,
+                                                 int iters) {
+#if K3_HAS_MFMA
+  i32x8 a, b;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    a[j] = (int)(0x3c3c3c3c + threadIdx.x + j);
+    b[j] = (int)(0x35353535 + threadIdx.x * 3 + j);
+  }
+  const int scale = 0x7f7f7f7f;  // E8M0 1.0 in every byte
+  f32x4 acc0 = {}, acc1 = {}, acc2 = {}, acc3 = {};
+  for (int i = 0; i < iters; ++i) {
+    acc0 = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+        a, b, acc0, FMT, FMT, 0, scale, 0, scale);
+    acc1 = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+        a, b, acc1, FMT, FMT, 0, scale, 0, scale);
+    acc2 = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+        a, b, acc2, FMT, FMT, 0, scale, 0, scale);
+    acc3 = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+        a, b, acc3, FMT, FMT, 0, scale, 0, scale);
+  }
+  float r = acc0[0] + acc1[1] + acc2[2] + acc3[3];
+  if (threadIdx.x == 0) out[blockIdx.x] = r;
+#else
+  if (threadIdx.x == 0) out[blockIdx.x] = -1.f;
+#endif
+}
+
 // Single-tile D[16x16] = A[16x32] * B[32x16] through one
 // v_mfma_f32_16x16x32_bf16 for numerics validation.
 //   layout 0 (validated on MI355X): lane l holds A[l&15][(l>>4)*8 + j]
