@@ -120,22 +120,34 @@ double mfma_throughput(torch::Tensor out, int64_t iters, int64_t shape) {
   K3_CHECK(out.is_cuda() && out.scalar_type() == torch::kFloat32 &&
                out.is_contiguous(),
            "out must be contiguous float32 GPU tensor");
-  K3_CHECK(shape == 16 || shape == 32, "shape must be 16 or 32");
+  K3_CHECK(shape == 16 || shape == 32 || shape == 8 || shape == 4,
+           "shape must be 16, 32 (bf16), 8 (MX-fp8) or 4 (MX-fp4)");
   int64_t blocks = out.numel();
   K3_CHECK(blocks > 0 && blocks <= (1 << 22), "bad block count");
   auto stream = at::hip::getCurrentHIPStream();
+  double flops_per_wave_iter = 4.0 * 16384.0;  // shapes 16/32 issue equal FLOPs
   if (shape == 16) {
     hipLaunchKernelGGL(k3samd_kern::mfma_throughput_kernel, dim3(blocks),
                        dim3(kThreadsPerBlock), 0, stream,
                        out.data_ptr<float>(), (int)iters);
-  } else {
+  } else if (shape == 32) {
     hipLaunchKernelGGL(k3samd_kern::mfma_throughput32_kernel, dim3(blocks),
                        dim3(kThreadsPerBlock), 0, stream,
                        out.data_ptr<float>(), (int)iters);
+  } else {
+    flops_per_wave_iter = 4.0 * 65536.0;  // K=128 MX shapes
+    if (shape == 8) {
+      hipLaunchKernelGGL(k3samd_kern::mfma_throughput_mx_kernel<0>,
+                         dim3(blocks), dim3(kThreadsPerBlock), 0, stream,
+                         out.data_ptr<float>(), (int)iters);
+    } else {
+      hipLaunchKernelGGL(k3samd_kern::mfma_throughput_mx_kernel<4>,
+                         dim3(blocks), dim3(kThreadsPerBlock), 0, stream,
+                         out.data_ptr<float>(), (int)iters);
+    }
   }
   C10_HIP_KERNEL_LAUNCH_CHECK();
-  // waves/block = 4; (4 acc x 16384 FLOP) or (2 acc x 32768 FLOP) per iter
-  return (double)blocks * 4.0 * 4.0 * 16384.0 * (double)iters;
+  return (double)blocks * 4.0 * flops_per_wave_iter * (double)iters;
 }
 
 torch::Tensor mfma_gemm16(torch::Tensor A, torch::Tensor B, int64_t layout) {

@@ -85,7 +85,7 @@ def test_mfma_gemm16_vs_torch():
 
 
 @requires_gpu
-@pytest.mark.parametrize("shape", [16, 32])
+@pytest.mark.parametrize("shape", [16, 32, 8, 4])
 def test_mfma_throughput_runs(shape):
     dev = torch.device("cuda", 0)
     out = torch.zeros(2048, device=dev)
