@@ -81,6 +81,19 @@ def test_human_table(native_bin, tmp_path):
     assert "256" in out
 
 
+def test_no_drm_tree(native_bin, tmp_path):
+    '''KFD present but /sys/class/drm absent (minimal initramfs-style
+    environments): enumeration still works, card unresolved, name falls
+    back.'''
+    root = build_tree(tmp_path / "sys", n_gpus=2, with_drm_cards=False)
+    out = json.loads(run_smi(native_bin, root, "--json"))
+    assert out["gpu_count"] == 2
+    for g in out["gpus"]:
+        assert g["card_index"] == -1
+        assert g["render_minor"] >= 128
+        assert g["name"]  # never empty
+
+
 def test_unique_id_fallback_to_bdf(native_bin, tmp_path):
     root = build_tree(tmp_path / "sys", n_gpus=1)
     # remove unique_id from properties
