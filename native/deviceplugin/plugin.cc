@@ -334,9 +334,33 @@ GrpcStatus DevicePlugin::handle_list_and_watch(
 
 std::string DevicePlugin::render_metrics() {
   size_t healthy = 0, unhealthy = 0;
+  std::vector<GpuDevice> gpus;
   {
     std::lock_guard<std::recursive_mutex> lk(mu_);
     for (const auto& vd : devices_) (vd.healthy ? healthy : unhealthy)++;
+    gpus = topo_.gpus;
+  }
+  // per-GPU runtime gauges (node-exporter style, read fresh per scrape)
+  std::string per_gpu =
+      "# TYPE k3samd_gpu_busy_percent gauge\n"
+      "# TYPE k3samd_gpu_temp_celsius gauge\n"
+      "# TYPE k3samd_gpu_power_watts gauge\n"
+      "# TYPE k3samd_gpu_vram_used_bytes gauge\n";
+  for (const auto& g : gpus) {
+    GpuRuntimeStats st = read_runtime_stats(sysfs_root_, g.card_index);
+    char line[512];
+    std::snprintf(line, sizeof(line),
+                  "k3samd_gpu_busy_percent{gpu=\"%s\"} %ld\n"
+                  "k3samd_gpu_temp_celsius{gpu=\"%s\"} %.1f\n"
+                  "k3samd_gpu_power_watts{gpu=\"%s\"} %.1f\n"
+                  "k3samd_gpu_vram_used_bytes{gpu=\"%s\"} %llu\n",
+                  g.stable_id().c_str(), st.busy_percent,
+                  g.stable_id().c_str(),
+                  st.temp_mc < 0 ? -1.0 : st.temp_mc / 1000.0,
+                  g.stable_id().c_str(),
+                  st.power_uw < 0 ? -1.0 : st.power_uw / 1e6,
+                  g.stable_id().c_str(), (unsigned long long)st.vram_used);
+    per_gpu += line;
   }
   char buf[2048];
   std::snprintf(
@@ -364,7 +388,7 @@ std::string DevicePlugin::render_metrics() {
       (unsigned long long)metrics_.list_and_watch_updates_total.load(),
       (unsigned long long)metrics_.registrations_total.load(),
       (unsigned long long)metrics_.health_transitions_total.load());
-  return buf;
+  return std::string(buf) + per_gpu;
 }
 
 bool DevicePlugin::serve(const std::string& plugin_sock,

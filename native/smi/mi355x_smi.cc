@@ -24,58 +24,10 @@
 
 namespace {
 
-std::string read_trim(const std::string& path) {
-  std::ifstream f(path);
-  if (!f) return {};
-  std::stringstream ss;
-  ss << f.rdbuf();
-  std::string s = ss.str();
-  while (!s.empty() && (s.back() == '\n' || s.back() == ' ')) s.pop_back();
-  return s;
-}
-
-uint64_t read_u64(const std::string& path, uint64_t dflt = 0) {
-  std::string s = read_trim(path);
-  if (s.empty()) return dflt;
-  return std::strtoull(s.c_str(), nullptr, 10);
-}
-
-struct CardStats {
-  uint64_t vram_used = 0;
-  uint64_t vram_total = 0;
-  long busy_percent = -1;
-  long temp_mc = -1;   // millidegrees C
-  long power_uw = -1;  // microwatts
-};
+using CardStats = k3samd::GpuRuntimeStats;
 
 CardStats card_stats(const std::string& sysfs_root, int card_index) {
-  CardStats st;
-  if (card_index < 0) return st;
-  std::string dev = sysfs_root + "/class/drm/card" +
-                    std::to_string(card_index) + "/device";
-  st.vram_used = read_u64(dev + "/mem_info_vram_used");
-  st.vram_total = read_u64(dev + "/mem_info_vram_total");
-  std::string busy = read_trim(dev + "/gpu_busy_percent");
-  if (!busy.empty()) st.busy_percent = std::strtol(busy.c_str(), nullptr, 10);
-  // scan hwmon/* (the index is global, not per-card, on real systems)
-  std::string hwdir = dev + "/hwmon";
-  if (DIR* d = ::opendir(hwdir.c_str())) {
-    while (struct dirent* e = ::readdir(d)) {
-      if (e->d_name[0] == '.') continue;
-      std::string base = hwdir + "/" + e->d_name;
-      std::string t = read_trim(base + "/temp1_input");
-      if (!t.empty() && st.temp_mc < 0)
-        st.temp_mc = std::strtol(t.c_str(), nullptr, 10);
-      // amdgpu exposes average or instantaneous package power
-      for (const char* f : {"/power1_average", "/power1_input"}) {
-        std::string pwr = read_trim(base + f);
-        if (!pwr.empty() && st.power_uw < 0)
-          st.power_uw = std::strtol(pwr.c_str(), nullptr, 10);
-      }
-    }
-    ::closedir(d);
-  }
-  return st;
+  return k3samd::read_runtime_stats(sysfs_root, card_index);
 }
 
 double mib(uint64_t b) { return (double)b / (1024.0 * 1024.0); }

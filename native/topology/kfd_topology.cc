@@ -94,6 +94,37 @@ std::map<std::string, uint64_t> parse_properties(const std::string& path) {
   return kv;
 }
 
+GpuRuntimeStats read_runtime_stats(const std::string& sysfs_root,
+                                   int card_index) {
+  GpuRuntimeStats st;
+  if (card_index < 0) return st;
+  const std::string dev =
+      sysfs_root + "/class/drm/card" + std::to_string(card_index) + "/device";
+  auto read_long = [](const std::string& p, long dflt) {
+    std::string v = trim(read_file(p));
+    return v.empty() ? dflt : std::strtol(v.c_str(), nullptr, 10);
+  };
+  st.vram_used = (uint64_t)read_long(dev + "/mem_info_vram_used", 0);
+  st.vram_total = (uint64_t)read_long(dev + "/mem_info_vram_total", 0);
+  st.busy_percent = read_long(dev + "/gpu_busy_percent", -1);
+  // hwmon index is global; scan the card's hwmon dir
+  const std::string hwdir = dev + "/hwmon";
+  if (DIR* d = ::opendir(hwdir.c_str())) {
+    while (struct dirent* e = ::readdir(d)) {
+      if (e->d_name[0] == '.') continue;
+      std::string base = hwdir + "/" + e->d_name;
+      if (st.temp_mc < 0) st.temp_mc = read_long(base + "/temp1_input", -1);
+      if (st.power_uw < 0) {
+        st.power_uw = read_long(base + "/power1_average", -1);
+        if (st.power_uw < 0)
+          st.power_uw = read_long(base + "/power1_input", -1);
+      }
+    }
+    ::closedir(d);
+  }
+  return st;
+}
+
 std::string default_sysfs_root() {
   const char* env = std::getenv("K3SAMD_SYSFS_ROOT");
   return env && *env ? env : "/sys";

@@ -68,4 +68,17 @@ Topology enumerate_topology(const std::string& sysfs_root = "/sys");
 // Resolve the sysfs root from the environment (K3SAMD_SYSFS_ROOT) or "/sys".
 std::string default_sysfs_root();
 
+// Per-GPU runtime stats from the amdgpu DRM sysfs (all best-effort; -1 /
+// 0 when a file is absent). Shared by mi355x-smi and the device plugin's
+// metrics endpoint.
+struct GpuRuntimeStats {
+  uint64_t vram_used = 0;   // bytes
+  uint64_t vram_total = 0;  // bytes (sysfs view; may differ from KFD heaps)
+  long busy_percent = -1;
+  long temp_mc = -1;        // millidegrees C
+  long power_uw = -1;       // microwatts
+};
+GpuRuntimeStats read_runtime_stats(const std::string& sysfs_root,
+                                   int card_index);
+
 }  // namespace k3samd

@@ -63,6 +63,12 @@ def test_metrics_endpoint(tmp_path):
         m = parse_metrics(body)
         assert m['k3samd_gpu_devices{health="healthy"}'] == 2
         assert m["k3samd_allocations_total"] == 0
+        # per-GPU runtime gauges from the fixture's DRM tree
+        gpu0 = 'gpu="amdgpu-1a2b3c4d5e6f0000"'
+        assert m[f"k3samd_gpu_busy_percent{{{gpu0}}}"] == 0
+        assert m[f"k3samd_gpu_temp_celsius{{{gpu0}}}"] == 53.0
+        assert m[f"k3samd_gpu_power_watts{{{gpu0}}}"] == 135.0
+        assert m[f"k3samd_gpu_vram_used_bytes{{{gpu0}}}"] == 2 * (1 << 20)
 
         # drive one allocation through grpc and re-scrape
         import grpc
