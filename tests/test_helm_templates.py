@@ -37,6 +37,7 @@ def test_daemonset_renders():
     assert c["image"] == "ghcr.io/k3samd/k3samd:0.1.0"
     assert "--config=/etc/k3samd/config.yaml" in c["command"]
     assert "--metrics-addr=0.0.0.0:9400" in c["command"]  # metrics default on
+    assert c["livenessProbe"]["httpGet"]["path"] == "/metrics"
     assert "--use-cdi" not in c["command"]  # cdi default off
     vols = {v["name"] for v in tmpl["volumes"]}
     assert {"device-plugins", "config", "sys"} <= vols
