@@ -361,7 +361,15 @@ void GrpcServer::serve_loop() {
       std::lock_guard<std::mutex> lk(conns_mu_);
       conns_.insert(conn);
     }
-    spawn([this, conn] {
+    spawn([this, conn] { connection_loop(conn); });
+  }
+}
+
+// One HTTP/2 connection, start to finish: preface handshake, frame loop,
+// RPC dispatch. Runs on its own (tracked) thread; returns when the peer
+// hangs up, errors, or the server stops.
+void GrpcServer::connection_loop(const std::shared_ptr<H2Conn>& conn) {
+  {  // scope: everything before the common close-and-deregister epilogue
       // --- connection handshake ---
       char preface[kPrefaceLen];
       if (!read_full(conn->fd, preface, kPrefaceLen, 10000) ||
@@ -594,13 +602,12 @@ void GrpcServer::serve_loop() {
           }
         }
       }
-    conn_done:
-      conn->close_fd();
-      {
-        std::lock_guard<std::mutex> lk(conns_mu_);
-        conns_.erase(conn);
-      }
-    });
+  }
+conn_done:
+  conn->close_fd();
+  {
+    std::lock_guard<std::mutex> lk(conns_mu_);
+    conns_.erase(conn);
   }
 }
 

@@ -60,6 +60,7 @@ class GrpcServer {
 
  private:
   void serve_loop();
+  void connection_loop(const std::shared_ptr<struct H2Conn>& conn);
 
   std::map<std::string, UnaryHandler> unary_;
   std::map<std::string, StreamHandler> stream_;
