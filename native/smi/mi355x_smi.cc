@@ -34,17 +34,37 @@ double mib(uint64_t b) { return (double)b / (1024.0 * 1024.0); }
 
 }  // namespace
 
+static int print_report(bool json);
+
 int main(int argc, char** argv) {
   if (k3samd::handle_version_flag(argc, argv, "mi355x-smi")) return 0;
   bool json = false;
+  int watch_s = 0;
   for (int i = 1; i < argc; ++i) {
     if (!std::strcmp(argv[i], "--json")) json = true;
+    else if ((!std::strcmp(argv[i], "--watch") || !std::strcmp(argv[i], "-l"))
+             && i + 1 < argc)
+      watch_s = std::atoi(argv[++i]);
     else if (!std::strcmp(argv[i], "--help") || !std::strcmp(argv[i], "-h")) {
-      std::printf("usage: mi355x-smi [--json]\n");
+      std::printf("usage: mi355x-smi [--json] [--watch SECONDS]\n");
       return 0;
     }
   }
+  if (watch_s > 0) {
+    // refresh loop (the nvidia-smi -l analog); re-exec the table printer
+    for (;;) {
+      std::printf("\033[2J\033[H");
+      int rc = print_report(json);
+      if (rc != 0) return rc;
+      std::fflush(stdout);
+      struct timespec ts = {watch_s, 0};
+      nanosleep(&ts, nullptr);
+    }
+  }
+  return print_report(json);
+}
 
+static int print_report(bool json) {
   const std::string root = k3samd::default_sysfs_root();
   k3samd::Topology topo = k3samd::enumerate_topology(root);
 
