@@ -17,3 +17,12 @@ def pytest_configure(config):
 @pytest.fixture
 def repo_root() -> Path:
     return REPO_ROOT
+
+
+@pytest.fixture(scope="session", autouse=True)
+def build_native():
+    """Every suite needs the native tree once per session (no-op when
+    up to date)."""
+    import subprocess
+    subprocess.run(["make", "-C", str(REPO_ROOT / "native"), "-j8"],
+                   check=True, capture_output=True)

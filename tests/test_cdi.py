@@ -16,12 +16,6 @@ BIN = REPO / "native" / "bin"
 IDENT = lambda b: b  # noqa: E731
 
 
-@pytest.fixture(scope="session", autouse=True)
-def build_native():
-    subprocess.run(["make", "-C", str(REPO / "native"), "-j8"], check=True,
-                   capture_output=True)
-
-
 def test_cdi_spec_generation(tmp_path):
     root = build_tree(tmp_path / "sys", n_gpus=2)
     out = subprocess.run(

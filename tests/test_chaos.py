@@ -21,12 +21,6 @@ PREFACE = b"PRI * HTTP/2.0\r\n\r\nSM\r\n\r\n"
 IDENT = lambda b: b  # noqa: E731
 
 
-@pytest.fixture(scope="session", autouse=True)
-def build_native():
-    subprocess.run(["make", "-C", str(REPO / "native"), "-j8"], check=True,
-                   capture_output=True)
-
-
 @pytest.fixture
 def plugin(tmp_path):
     root = build_tree(tmp_path / "sys", n_gpus=1)

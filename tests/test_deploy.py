@@ -15,12 +15,6 @@ CHART = REPO / "deploy" / "helm" / "k3samd-device-plugin"
 PLUGIN = REPO / "native" / "bin" / "k3samd-device-plugin"
 
 
-@pytest.fixture(scope="session", autouse=True)
-def build_native():
-    subprocess.run(["make", "-C", str(REPO / "native"), "-j8"], check=True,
-                   capture_output=True)
-
-
 def load_docs(path):
     return [d for d in yaml.safe_load_all(path.read_text()) if d]
 

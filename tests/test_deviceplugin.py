@@ -28,12 +28,6 @@ PLUGIN = REPO / "native" / "bin" / "k3samd-device-plugin"
 IDENT = lambda b: b  # noqa: E731  identity (de)serializer
 
 
-@pytest.fixture(scope="session", autouse=True)
-def build_native():
-    subprocess.run(["make", "-C", str(REPO / "native"), "-j8"], check=True,
-                   capture_output=True)
-
-
 class FakeKubelet:
     """Records Register calls, like kubelet's Registration service."""
 

@@ -22,12 +22,6 @@ REPO = Path(__file__).resolve().parent.parent
 RUNTIME = REPO / "native" / "bin" / "k3samd-oci-runtime"
 
 
-@pytest.fixture(scope="session", autouse=True)
-def build_native():
-    subprocess.run(["make", "-C", str(REPO / "native"), "-j8"], check=True,
-                   capture_output=True)
-
-
 def oci_transform(tmp_path, sysfs_root, envs):
     spec = {
         "ociVersion": "1.0.2",

@@ -12,12 +12,6 @@ REPO = Path(__file__).resolve().parent.parent
 SELFTEST = REPO / "native" / "bin" / "k3samd-selftest"
 
 
-@pytest.fixture(scope="session", autouse=True)
-def build_native():
-    subprocess.run(["make", "-C", str(REPO / "native"), "-j8"], check=True,
-                   capture_output=True)
-
-
 def run_fuzz(mode: str, data: bytes):
     proc = subprocess.run([str(SELFTEST), "--fuzz", mode], input=data,
                           capture_output=True, timeout=60)

@@ -22,12 +22,6 @@ HEADERS, DATA, SETTINGS, CONTINUATION, WINDOW_UPDATE = 0x1, 0x0, 0x4, 0x9, 0x8
 END_STREAM, ACK, END_HEADERS, PADDED = 0x1, 0x1, 0x4, 0x8
 
 
-@pytest.fixture(scope="session", autouse=True)
-def build_native():
-    subprocess.run(["make", "-C", str(REPO / "native"), "-j8"], check=True,
-                   capture_output=True)
-
-
 def frame(ftype, flags, stream, payload):
     return struct.pack(">I", len(payload))[1:] + bytes([ftype, flags]) + \
         struct.pack(">I", stream) + payload

@@ -9,12 +9,6 @@ import pytest
 REPO = Path(__file__).resolve().parent.parent
 
 
-@pytest.fixture(scope="session", autouse=True)
-def build_native():
-    subprocess.run(["make", "-C", str(REPO / "native"), "-j8"], check=True,
-                   capture_output=True)
-
-
 def test_install_to_prefix(tmp_path):
     prefix = tmp_path / "bin"
     prefix.mkdir()

@@ -14,12 +14,6 @@ REPO = Path(__file__).resolve().parent.parent
 LABELLER = REPO / "native" / "bin" / "k3samd-node-labeller"
 
 
-@pytest.fixture(scope="session", autouse=True)
-def build_native():
-    subprocess.run(["make", "-C", str(REPO / "native"), "-j8"], check=True,
-                   capture_output=True)
-
-
 def run_labeller(sysfs_root, *args):
     return subprocess.run(
         [str(LABELLER), *args],

@@ -16,12 +16,6 @@ PLUGIN = REPO / "native" / "bin" / "k3samd-device-plugin"
 IDENT = lambda b: b  # noqa: E731
 
 
-@pytest.fixture(scope="session", autouse=True)
-def build_native():
-    subprocess.run(["make", "-C", str(REPO / "native"), "-j8"], check=True,
-                   capture_output=True)
-
-
 def scrape(metrics_sock):
     s = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
     s.connect(metrics_sock)

@@ -18,12 +18,6 @@ BIN = REPO / "native" / "bin"
 IDENT = lambda b: b  # noqa: E731
 
 
-@pytest.fixture(scope="session", autouse=True)
-def build_native():
-    subprocess.run(["make", "-C", str(REPO / "native"), "-j8"], check=True,
-                   capture_output=True)
-
-
 def test_selftest_passes():
     proc = subprocess.run([str(BIN / "k3samd-selftest")],
                           capture_output=True, text=True, timeout=120)

@@ -19,12 +19,6 @@ REPO = Path(__file__).resolve().parent.parent
 SMI = REPO / "native" / "bin" / "mi355x-smi"
 
 
-@pytest.fixture(scope="session", autouse=True)
-def build_native():
-    subprocess.run(["make", "-C", str(REPO / "native"), "-j8"], check=True,
-                   capture_output=True)
-
-
 def make_partitioned_tree(root):
     """2 partitions of one physical GPU: same BDF/card, render 128/129."""
     root = build_tree(root, n_gpus=1, n_cpu_nodes=1)

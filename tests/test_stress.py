@@ -17,12 +17,6 @@ IDENT = lambda b: b  # noqa: E731
 REPO = Path(__file__).resolve().parent.parent
 
 
-@pytest.fixture(scope="session", autouse=True)
-def build_native():
-    subprocess.run(["make", "-C", str(REPO / "native"), "-j8"], check=True,
-                   capture_output=True)
-
-
 def fd_count(pid):
     return len(os.listdir(f"/proc/{pid}/fd"))
 
