@@ -275,12 +275,14 @@ int main(int argc, char** argv) {
     mfma32_tf = time_mfma(k3samd_kern::mfma_throughput32_kernel, 32768.0, 2);
     double mx8_tf =
         time_mfma(k3samd_kern::mfma_throughput_mx_kernel<0>, 65536.0, 4);
+    double mx6_tf =
+        time_mfma(k3samd_kern::mfma_throughput_mx_kernel<2>, 65536.0, 4);
     double mx4_tf =
         time_mfma(k3samd_kern::mfma_throughput_mx_kernel<4>, 65536.0, 4);
     std::printf("| MFMA bf16 16x16x32: %7.1f TF | 32x32x16: %7.1f TF     |\n",
                 mfma_tf, mfma32_tf);
-    std::printf("| MFMA MX-fp8 16x16x128: %6.1f TF | MX-fp4: %7.1f TF    |\n",
-                mx8_tf, mx4_tf);
+    std::printf("| MFMA MX 16x16x128  fp8: %6.1f | fp6: %6.1f | fp4: %6.1f |\n",
+                mx8_tf, mx6_tf, mx4_tf);
     std::printf("+--------+-------------------+-------------------+\n");
     if (mfma32_tf > mfma_tf) mfma_tf = mfma32_tf;
     HIP_CHECK(hipFree(out));
