@@ -81,7 +81,12 @@ def main(argv=None) -> int:
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
-    n_gpus = args.gpus or world_size
+    # the aggregate is computed from ranks that actually ran: WORLD_SIZE is
+    # the source of truth; --gpus is a cross-check only (never an inflator)
+    n_gpus = world_size
+    if args.gpus is not None and args.gpus != world_size:
+        print(f"# note: --gpus {args.gpus} != WORLD_SIZE {world_size}; "
+              f"reporting n_gpus={world_size}", file=sys.stderr)
 
     if not torch.cuda.is_available():
         print("bench.py requires a GPU (run on the MI355X box)",
