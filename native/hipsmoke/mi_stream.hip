@@ -18,6 +18,7 @@
 #include <vector>
 
 #include "../../k3samd/ops/hip/stream_kernels.h"
+#include "../topology/kfd_topology.h"
 
 #define HIP_CHECK(x)                                                         \
   do {                                                                       \
@@ -48,6 +49,7 @@ int main(int argc, char** argv) {
   bool do_mfma = true;
   bool tune = false;
   bool all_gpus = false;
+  int burn_s = 0;
   for (int i = 1; i < argc; ++i) {
     if (!std::strcmp(argv[i], "--mib") && i + 1 < argc)
       mib = std::atoll(argv[++i]);
@@ -61,9 +63,11 @@ int main(int argc, char** argv) {
       tune = true;
     else if (!std::strcmp(argv[i], "--all-gpus"))
       all_gpus = true;
+    else if (!std::strcmp(argv[i], "--burn") && i + 1 < argc)
+      burn_s = std::atoi(argv[++i]);
     else {
       std::printf("mi-stream [--mib N] [--iters N] [--device D] [--no-mfma]"
-                  " [--tune] [--all-gpus]\n");
+                  " [--tune] [--all-gpus] [--burn SECONDS]\n");
       return !std::strcmp(argv[i], "--help") ? 0 : 2;
     }
   }
@@ -190,6 +194,84 @@ int main(int argc, char** argv) {
                     blocks, g1, g2, g4);
       }
     }
+    (void)hipFree(a); (void)hipFree(b); (void)hipFree(c);
+    return 0;
+  }
+
+  if (burn_s > 0) {
+    // Burn-in: saturate HBM (nt triad) and the matrix pipes (bf16 MFMA)
+    // concurrently on two streams for --burn seconds, sampling thermals
+    // from sysfs. Node-acceptance analog of gpu-burn.
+    hipStream_t s_mem, s_mfma;
+    HIP_CHECK(hipStreamCreate(&s_mem));
+    HIP_CHECK(hipStreamCreate(&s_mfma));
+    float* mf_out;
+    const int mf_blocks = 1024;
+    HIP_CHECK(hipMalloc(&mf_out, mf_blocks * sizeof(float)));
+    // resolve our card index for sysfs sampling (best effort)
+    int card = -1;
+    {
+      auto topo = k3samd::enumerate_topology(k3samd::default_sysfs_root());
+      if (!topo.gpus.empty()) card = topo.gpus[0].card_index;
+    }
+    auto now = [] { return std::chrono::steady_clock::now(); };
+    auto t_end = now() + std::chrono::seconds(burn_s);
+    uint64_t triads = 0, mfmas = 0;
+    long max_temp = -1, max_pw = -1;
+    std::printf("burn: %d s of concurrent HBM streaming + bf16 MFMA\n",
+                burn_s);
+    std::printf("%6s %12s %12s %6s %7s %6s\n", "t(s)", "GB/s", "TFLOP/s",
+                "temp", "power", "busy");
+    auto t0 = now();
+    auto next_report = t0 + std::chrono::seconds(2);
+    uint64_t triads_last = 0, mfmas_last = 0;
+    auto t_last = t0;
+    while (now() < t_end) {
+      for (int i = 0; i < 8; ++i) {
+        hipLaunchKernelGGL(k3samd_kern::stream_triad_kernel<true>, grid,
+                           block, 0, s_mem, a, b, c, s, n4);
+        hipLaunchKernelGGL(k3samd_kern::mfma_throughput_kernel,
+                           dim3(mf_blocks), block, 0, s_mfma, mf_out, 512);
+        ++triads;
+        ++mfmas;
+      }
+      HIP_CHECK(hipStreamSynchronize(s_mem));
+      HIP_CHECK(hipStreamSynchronize(s_mfma));
+      if (now() >= next_report) {
+        auto st = k3samd::read_runtime_stats(k3samd::default_sysfs_root(),
+                                             card);
+        double dt = std::chrono::duration<double>(now() - t_last).count();
+        double gbs = (triads - triads_last) * 3.0 * buf_bytes / dt / 1e9;
+        double tf =
+            (mfmas - mfmas_last) * mf_blocks * 4.0 * 4.0 * 16384.0 * 512.0 /
+            dt / 1e12;
+        if (st.temp_mc > max_temp) max_temp = st.temp_mc;
+        if (st.power_uw > max_pw) max_pw = st.power_uw;
+        std::printf("%6.0f %12.1f %12.1f %5ldC %6.0fW %5ld%%\n",
+                    std::chrono::duration<double>(now() - t0).count(), gbs,
+                    tf, st.temp_mc < 0 ? -1 : st.temp_mc / 1000,
+                    st.power_uw < 0 ? -1.0 : st.power_uw / 1e6,
+                    st.busy_percent);
+        std::fflush(stdout);
+        triads_last = triads;
+        mfmas_last = mfmas;
+        t_last = now();
+        next_report += std::chrono::seconds(2);
+      }
+    }
+    double total_s = std::chrono::duration<double>(now() - t0).count();
+    double avg_gbs = triads * 3.0 * buf_bytes / total_s / 1e9;
+    double avg_tf =
+        mfmas * mf_blocks * 4.0 * 4.0 * 16384.0 * 512.0 / total_s / 1e12;
+    std::printf(
+        "{\"payload\": \"mi-burn\", \"seconds\": %.0f, \"avg_triad_gbps\": "
+        "%.1f, \"avg_mfma_tflops\": %.1f, \"max_temp_c\": %ld, "
+        "\"max_power_w\": %.0f}\n",
+        total_s, avg_gbs, avg_tf, max_temp < 0 ? -1 : max_temp / 1000,
+        max_pw < 0 ? -1.0 : max_pw / 1e6);
+    HIP_CHECK(hipFree(mf_out));
+    HIP_CHECK(hipStreamDestroy(s_mem));
+    HIP_CHECK(hipStreamDestroy(s_mfma));
     (void)hipFree(a); (void)hipFree(b); (void)hipFree(c);
     return 0;
   }
