@@ -47,6 +47,18 @@ def test_mi_stream_all_gpus():
 
 
 @requires_gpu
+def test_mi_stream_burn_short():
+    proc = subprocess.run([str(BIN / "mi-stream"), "--burn", "4",
+                           "--mib", "256"],
+                          capture_output=True, text=True, timeout=300)
+    assert proc.returncode == 0, proc.stderr
+    j = json.loads(proc.stdout.strip().splitlines()[-1])
+    assert j["payload"] == "mi-burn"
+    assert j["avg_triad_gbps"] > 2000  # concurrent with MFMA load
+    assert j["avg_mfma_tflops"] > 50
+
+
+@requires_gpu
 def test_mi_allreduce_single_gpu():
     proc = subprocess.run([str(BIN / "mi-allreduce"), "--ngpus", "1",
                            "--min-mib", "4", "--max-mib", "16",
