@@ -38,9 +38,16 @@ def summarize(db_path: str | Path) -> str:
 
     lines.append("| kernel | calls | total ms | avg us | % GPU time |")
     lines.append("|---|---|---|---|---|")
-    for name, calls, total_us, avg_us, pct in cur.execute(
+    try:
+        rows = list(cur.execute(
             "SELECT name, total_calls, total_duration, average, percentage "
-            "FROM top_kernels ORDER BY percentage DESC LIMIT 15"):
+            "FROM top_kernels ORDER BY percentage DESC LIMIT 15"))
+    except sqlite3.Error as e:
+        con.close()
+        raise SystemExit(
+            f"{db_path}: no top_kernels view — was this db produced by "
+            f"rocprofv3 --kernel-trace --stats? ({e})")
+    for name, calls, total_us, avg_us, pct in rows:
         lines.append(f"| `{shorten(name)}` | {calls} | "
                      f"{total_us / 1e3:.2f} | {avg_us:.1f} | {pct:.1f} |")
     con.close()
