@@ -346,6 +346,20 @@ def test_reregister_after_kubelet_restart(harness):
     assert h.kubelet.requests[0]["resource_name"] == "amd.com/gpu"
 
 
+def test_register_failure_exits_nonzero(tmp_path):
+    '''No kubelet at the socket => serve() fails and the daemon exits 1
+    (DaemonSet backoff semantics, not a silent half-start).'''
+    root = build_tree(tmp_path / "sys", n_gpus=1)
+    proc = subprocess.run(
+        [str(PLUGIN), "--plugin-sock", str(tmp_path / "amd.sock"),
+         "--kubelet-sock", str(tmp_path / "missing-kubelet.sock")],
+        env={"K3SAMD_SYSFS_ROOT": str(root)},
+        capture_output=True, text=True, timeout=60)
+    assert proc.returncode == 1
+    assert "Register with kubelet failed" in proc.stderr
+    assert not (tmp_path / "amd.sock").exists()  # socket cleaned up
+
+
 def test_cpu_only_zero_allocatable(tmp_path):
     """BASELINE.json config #1: CPU-only node => 0 amd.com/gpu."""
     root = build_tree(tmp_path / "sys", n_gpus=0)

@@ -48,6 +48,26 @@ def test_features_file(tmp_path):
     assert not (tmp_path / "features.d" / "k3samd.tmp").exists()
 
 
+def test_daemon_mode_writes_then_terminates(tmp_path):
+    import signal
+    import time
+    root = build_tree(tmp_path / "sys", n_gpus=1)
+    ff = tmp_path / "k3samd"
+    proc = subprocess.Popen(
+        [str(LABELLER), "--features-file", str(ff), "--interval-s", "1"],
+        env={"K3SAMD_SYSFS_ROOT": str(root)})
+    try:
+        deadline = time.time() + 10
+        while not ff.exists():
+            assert time.time() < deadline and proc.poll() is None
+            time.sleep(0.05)
+        lines = dict(l.split("=", 1) for l in ff.read_text().splitlines())
+        assert lines["amd.com/gpu.present"] == "true"
+    finally:
+        proc.send_signal(signal.SIGTERM)
+        proc.wait(timeout=10)
+
+
 def test_cpu_only_no_labels(tmp_path):
     root = build_tree(tmp_path / "sys", n_gpus=0)
     ff = tmp_path / "k3samd"
