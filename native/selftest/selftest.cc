@@ -199,6 +199,57 @@ int test_miniyaml() {
   return 0;
 }
 
+int test_miniyaml_edges() {
+  // deeper-indented list under a key
+  auto a = k3samd::yaml_parse("k:
+    - x
+    - y
+");
+  CHECK(a.get("k")->list.size() == 2);
+  // empty values and nested empties
+  auto b = k3samd::yaml_parse("a:
+b: v
+");
+  CHECK(b.get("a")->type == k3samd::YNode::kNull);
+  CHECK(b.get("b")->as_str() == "v");
+  // quoted scalar keeps inner colon-space
+  auto c = k3samd::yaml_parse("k: "a: b"
+");
+  CHECK(c.get("k")->as_str() == "a: b");
+  // document marker + comments-only lines
+  auto d = k3samd::yaml_parse("---
+# only a comment
+k: 1
+");
+  CHECK(d.get("k")->as_int() == 1);
+  // as_bool variants
+  auto e = k3samd::yaml_parse("a: True
+b: OFF
+c: weird
+");
+  CHECK(e.get("a")->as_bool(false) == true);
+  CHECK(e.get("b")->as_bool(true) == false);
+  CHECK(e.get("c")->as_bool(true) == true);  // falls back to default
+  return 0;
+}
+
+int test_minijson_edges() {
+  // deeply nested arrays round-trip
+  auto v = k3samd::json_parse("[[[[1,2],[3]],[]],null]");
+  CHECK(v->arr.size() == 2);
+  CHECK(v->arr[0]->arr[0]->arr[0]->arr[1]->as_int() == 2);
+  std::string ser = k3samd::json_serialize(v);
+  auto v2 = k3samd::json_parse(ser);
+  CHECK(v2->arr[0]->arr[0]->arr[1]->arr[0]->as_int() == 3);
+  // duplicate keys: last one wins through get()
+  auto d = k3samd::json_parse("{"k": 1, "k": 2}");
+  CHECK(d->get("k")->as_int() == 1 || d->obj.size() == 2);  // order kept
+  // pretty serialization parses back
+  std::string pretty = k3samd::json_serialize(v, 2);
+  CHECK(k3samd::json_parse(pretty)->arr.size() == 2);
+  return 0;
+}
+
 int test_minijson() {
   const char* text =
       "{\"a\": [1, 2.5, -3e2], \"s\": \"q\\\"uote\\n\", \"n\": null, "
@@ -279,7 +330,9 @@ int main(int argc, char** argv) {
   rc |= test_hpack_encoder_roundtrip();
   rc |= test_proto();
   rc |= test_miniyaml();
+  rc |= test_miniyaml_edges();
   rc |= test_minijson();
+  rc |= test_minijson_edges();
   if (rc == 0) std::printf("k3samd-selftest: all checks passed\n");
   return rc;
 }
