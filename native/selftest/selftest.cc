@@ -201,32 +201,20 @@ int test_miniyaml() {
 
 int test_miniyaml_edges() {
   // deeper-indented list under a key
-  auto a = k3samd::yaml_parse("k:
-    - x
-    - y
-");
+  auto a = k3samd::yaml_parse("k:\n    - x\n    - y\n");
   CHECK(a.get("k")->list.size() == 2);
-  // empty values and nested empties
-  auto b = k3samd::yaml_parse("a:
-b: v
-");
+  // empty values next to filled ones
+  auto b = k3samd::yaml_parse("a:\nb: v\n");
   CHECK(b.get("a")->type == k3samd::YNode::kNull);
   CHECK(b.get("b")->as_str() == "v");
   // quoted scalar keeps inner colon-space
-  auto c = k3samd::yaml_parse("k: "a: b"
-");
+  auto c = k3samd::yaml_parse("k: \"a: b\"\n");
   CHECK(c.get("k")->as_str() == "a: b");
-  // document marker + comments-only lines
-  auto d = k3samd::yaml_parse("---
-# only a comment
-k: 1
-");
+  // document marker + comment-only lines
+  auto d = k3samd::yaml_parse("---\n# only a comment\nk: 1\n");
   CHECK(d.get("k")->as_int() == 1);
   // as_bool variants
-  auto e = k3samd::yaml_parse("a: True
-b: OFF
-c: weird
-");
+  auto e = k3samd::yaml_parse("a: True\nb: OFF\nc: weird\n");
   CHECK(e.get("a")->as_bool(false) == true);
   CHECK(e.get("b")->as_bool(true) == false);
   CHECK(e.get("c")->as_bool(true) == true);  // falls back to default
@@ -241,9 +229,9 @@ int test_minijson_edges() {
   std::string ser = k3samd::json_serialize(v);
   auto v2 = k3samd::json_parse(ser);
   CHECK(v2->arr[0]->arr[0]->arr[1]->arr[0]->as_int() == 3);
-  // duplicate keys: last one wins through get()
-  auto d = k3samd::json_parse("{"k": 1, "k": 2}");
-  CHECK(d->get("k")->as_int() == 1 || d->obj.size() == 2);  // order kept
+  // duplicate keys are preserved in order; get() returns the first
+  auto d = k3samd::json_parse("{\"k\": 1, \"k\": 2}");
+  CHECK(d->get("k")->as_int() == 1 && d->obj.size() == 2);
   // pretty serialization parses back
   std::string pretty = k3samd::json_serialize(v, 2);
   CHECK(k3samd::json_parse(pretty)->arr.size() == 2);
