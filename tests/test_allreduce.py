@@ -7,6 +7,7 @@ run at round end is correct by construction.
 
 import json
 import os
+import socket
 import subprocess
 import sys
 from pathlib import Path
@@ -14,6 +15,14 @@ from pathlib import Path
 import pytest
 
 REPO = Path(__file__).resolve().parent.parent
+
+
+def free_port() -> str:
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return str(port)
 
 WORKER = r"""
 import json, os, sys
@@ -53,11 +62,12 @@ def test_gloo_world2(tmp_path):
     script = tmp_path / "worker.py"
     script.write_text(WORKER)
     procs = []
+    port = free_port()
     for rank in range(2):
         env = dict(os.environ)
         env.update({
             "RANK": str(rank), "WORLD_SIZE": "2",
-            "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": "29781",
+            "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": port,
             "K3_REPO": str(REPO),
         })
         procs.append(subprocess.Popen(
@@ -71,7 +81,7 @@ def test_gloo_world2(tmp_path):
 
 def test_cli_single_rank():
     env = dict(os.environ)
-    env.update({"MASTER_PORT": "29782"})
+    env.update({"MASTER_ADDR": "127.0.0.1", "MASTER_PORT": free_port()})
     proc = subprocess.run(
         [sys.executable, "-m", "k3samd.parallel.allreduce",
          "--backend", "gloo", "--min-mib", "1", "--max-mib", "1",
