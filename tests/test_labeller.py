@@ -68,6 +68,21 @@ def test_daemon_mode_writes_then_terminates(tmp_path):
         proc.wait(timeout=10)
 
 
+def test_mixed_arch_uses_first_gpu(tmp_path):
+    '''Heterogeneous nodes are out of scope for MI355X clusters, but the
+    labeller must stay deterministic: labels describe the first GPU.'''
+    root = build_tree(tmp_path / "sys", n_gpus=2)
+    # mutate the second GPU's arch to gfx942
+    props = root / "class/kfd/kfd/topology/nodes/3/properties"
+    text = props.read_text().replace("gfx_target_version 90500",
+                                     "gfx_target_version 90402")
+    props.write_text(text)
+    out = run_labeller(root, "--json")
+    labels = json.loads(out.stdout)
+    assert labels["amd.com/gpu.arch"] == "gfx950"   # first GPU
+    assert labels["amd.com/gpu.count"] == "2"
+
+
 def test_cpu_only_no_labels(tmp_path):
     root = build_tree(tmp_path / "sys", n_gpus=0)
     ff = tmp_path / "k3samd"
