@@ -39,6 +39,44 @@ def parse_metrics(body: str) -> dict:
     return out
 
 
+def free_port():
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+def test_metrics_tcp_endpoint(tmp_path):
+    """The chart serves metrics over TCP (0.0.0.0:9400); cover that path."""
+    import urllib.request
+    root = build_tree(tmp_path / "sys", n_gpus=1)
+    sock = str(tmp_path / "amd.sock")
+    port = free_port()
+    proc = subprocess.Popen(
+        [str(PLUGIN), "--plugin-sock", sock, "--no-register",
+         "--metrics-addr", f"127.0.0.1:{port}", "--health-poll-ms", "0"],
+        env={"K3SAMD_SYSFS_ROOT": str(root)},
+        stdout=subprocess.PIPE, stderr=subprocess.PIPE)
+    try:
+        deadline = time.time() + 10
+        body = None
+        while time.time() < deadline and proc.poll() is None:
+            try:
+                body = urllib.request.urlopen(
+                    f"http://127.0.0.1:{port}/metrics", timeout=5
+                ).read().decode()
+                break
+            except OSError:
+                time.sleep(0.1)
+        assert body is not None, "metrics endpoint never came up"
+        m = parse_metrics(body)
+        assert m['k3samd_gpu_devices{health="healthy"}'] == 1
+    finally:
+        proc.terminate()
+        proc.wait(timeout=10)
+
+
 def test_metrics_endpoint(tmp_path):
     root = build_tree(tmp_path / "sys", n_gpus=2)
     sock = str(tmp_path / "amd.sock")
