@@ -346,6 +346,40 @@ def test_reregister_after_kubelet_restart(harness):
     assert h.kubelet.requests[0]["resource_name"] == "amd.com/gpu"
 
 
+def test_socket_takeover_on_restart(tmp_path):
+    '''A replacement plugin instance (DaemonSet restart with a stale pod
+    lingering) takes over the socket path; clients reach the new one.'''
+    root = build_tree(tmp_path / "sys", n_gpus=1)
+    sock = str(tmp_path / "amd.sock")
+    env = {"K3SAMD_SYSFS_ROOT": str(root)}
+    p1 = subprocess.Popen([str(PLUGIN), "--plugin-sock", sock,
+                           "--no-register"], env=env,
+                          stdout=subprocess.PIPE, stderr=subprocess.PIPE)
+    p2 = None
+    try:
+        deadline = time.time() + 10
+        while not Path(sock).exists():
+            assert time.time() < deadline and p1.poll() is None
+            time.sleep(0.02)
+        p2 = subprocess.Popen([str(PLUGIN), "--plugin-sock", sock,
+                               "--no-register"], env=env,
+                              stdout=subprocess.PIPE, stderr=subprocess.PIPE)
+        time.sleep(0.5)
+        assert p2.poll() is None, p2.stderr.read()[:300]
+        ch = grpc.insecure_channel(f"unix:{sock}")
+        opts = pb.decode_options(
+            ch.unary_unary("/v1beta1.DevicePlugin/GetDevicePluginOptions",
+                           request_serializer=IDENT,
+                           response_deserializer=IDENT)(b"", timeout=10))
+        assert opts["get_preferred_allocation_available"] is True
+        ch.close()
+    finally:
+        for p in (p1, p2):
+            if p is not None:
+                p.terminate()
+                p.wait(timeout=10)
+
+
 def test_register_failure_exits_nonzero(tmp_path):
     '''No kubelet at the socket => serve() fails and the daemon exits 1
     (DaemonSet backoff semantics, not a silent half-start).'''
