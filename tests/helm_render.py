@@ -72,7 +72,12 @@ def render(template: str, ctx: Ctx) -> str:
         pos = m.end()
         expr = m.group(1)
         if expr.startswith("if "):
-            val = lookup(ctx, expr[3:])
+            cond = expr[3:].split()
+            if cond and cond[0] in ("and", "or"):
+                vals = [bool(lookup(ctx, c)) for c in cond[1:]]
+                val = all(vals) if cond[0] == "and" else any(vals)
+            else:
+                val = bool(lookup(ctx, expr[3:]))
             stack.append(("if", bool(val), ctx.dot))
         elif expr.startswith("with "):
             val = lookup(ctx, expr[5:])

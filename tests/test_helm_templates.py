@@ -71,6 +71,16 @@ def test_configmap_embeds_plugin_config():
     assert cfg["sharing"]["timeSlicing"]["resources"][0]["replicas"] == 4
 
 
+def test_servicemonitor_gated():
+    assert render_template("servicemonitor.yaml") == []
+    docs = render_template("servicemonitor.yaml",
+                           {"metrics": {"serviceMonitor": {"enabled": True}}})
+    kinds = [d["kind"] for d in docs]
+    assert kinds == ["Service", "ServiceMonitor"]
+    sm = docs[1]
+    assert sm["spec"]["endpoints"][0]["port"] == "metrics"
+
+
 def test_runtimeclass_renders():
     (rc,) = render_template("runtimeclass.yaml")
     assert rc["kind"] == "RuntimeClass"
