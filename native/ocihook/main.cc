@@ -53,6 +53,8 @@ bool transform_config(const std::string& path) {
   if (const char* r = std::getenv("K3SAMD_ROCM_ROOT")) opts.rocm_root = r;
   if (const char* e = std::getenv("K3SAMD_INJECT_ROCM_DEFAULT"))
     opts.inject_rocm_default = !std::strcmp(e, "1");
+  if (const char* e = std::getenv("K3SAMD_ALLOW_ALL"))
+    opts.allow_all_default = !std::strcmp(e, "1");
 
   k3samd::Topology topo = k3samd::enumerate_topology(opts.sysfs_root);
   k3samd::InjectReport report;
@@ -71,7 +73,10 @@ bool transform_config(const std::string& path) {
   std::fprintf(stderr,
                "k3samd-oci-runtime: injected %zu device(s), %zu mount(s)%s\n",
                report.devices_added.size(), report.mounts_added.size(),
-               report.skipped ? " (skipped: visible=none)" : "");
+               report.no_allocation
+                   ? " (skipped: no allocation — default-deny; set "
+                     "K3SAMD_VISIBLE_DEVICES or K3SAMD_ALLOW_ALL=1)"
+                   : (report.skipped ? " (skipped: visible=none)" : ""));
   return true;
 }
 

@@ -144,10 +144,25 @@ bool oci_inject_gpus(const JPtr& config, const Topology& topo,
     return true;
   }
 
-  // pick GPUs: default all; else by stable id; render-minor list also
-  // honored (set by the device plugin alongside the id list)
+  // Default-deny: a container with NO allocation (no env, no annotation)
+  // gets nothing. Injecting every GPU here would let any pod that merely
+  // sets runtimeClassName bypass kubelet device accounting — the
+  // NVIDIA_VISIBLE_DEVICES=all foot-gun. Injecting all GPUs requires the
+  // explicit "all" sentinel, or the operator-level allow_all_default
+  // runtime flag (K3SAMD_ALLOW_ALL=1 on the runtime binary, debug only).
+  if (visible.empty() && minors_env.empty() && !opts.allow_all_default) {
+    if (report) {
+      report->skipped = true;
+      report->no_allocation = true;
+    }
+    return true;
+  }
+
+  // pick GPUs: "all" sentinel (or operator allow-all default) = every GPU;
+  // else by stable id; render-minor list also honored (set by the device
+  // plugin alongside the id list)
   std::vector<const GpuDevice*> selected;
-  if (visible.empty() && minors_env.empty()) {
+  if (visible == "all" || (visible.empty() && minors_env.empty())) {
     for (const auto& g : topo.gpus) selected.push_back(&g);
   } else {
     std::set<std::string> want_ids;

@@ -9,8 +9,13 @@
 // selected /dev/dri render+card nodes, matching device-cgroup allow rules,
 // and (optionally) the host ROCm userspace as read-only bind mounts.
 //
-// Device selection mirrors the NVIDIA env-var contract:
-//   K3SAMD_VISIBLE_DEVICES unset  -> all GPUs (RuntimeClass alone suffices)
+// Device selection is default-deny (unlike NVIDIA's implicit-all, which
+// lets any runtimeClassName pod bypass kubelet device accounting):
+//   K3SAMD_VISIBLE_DEVICES unset  -> NOTHING injected (no allocation ⇒ no
+//                                    GPUs), unless the operator set the
+//                                    runtime-level allow_all_default flag
+//                                    (K3SAMD_ALLOW_ALL=1, debug only)
+//   K3SAMD_VISIBLE_DEVICES=all    -> all GPUs (explicit opt-in sentinel)
 //   K3SAMD_VISIBLE_DEVICES=none   -> nothing injected
 //   K3SAMD_VISIBLE_DEVICES=<ids>  -> the listed stable ids (set by the
 //                                    device plugin's Allocate response)
@@ -30,12 +35,15 @@ struct InjectOptions {
   std::string dev_root = "/dev";
   std::string rocm_root = "/opt/rocm";  // host path bind-mounted when asked
   bool inject_rocm_default = false;     // K3SAMD_INJECT_ROCM=1 overrides
+  bool allow_all_default = false;       // K3SAMD_ALLOW_ALL=1 (runtime env,
+                                        // debug): no-allocation ⇒ all GPUs
 };
 
 struct InjectReport {
   std::vector<std::string> devices_added;
   std::vector<std::string> mounts_added;
-  bool skipped = false;  // K3SAMD_VISIBLE_DEVICES=none/void
+  bool skipped = false;        // nothing injected
+  bool no_allocation = false;  // skipped because no env/annotation present
 };
 
 // Mutates `config` (an OCI runtime spec DOM) in place.

@@ -71,7 +71,8 @@ def test_labeller_real_sysfs():
 def test_oci_transform_real_topology(tmp_path):
     spec = {
         "ociVersion": "1.0.2",
-        "process": {"args": ["true"], "env": [], "cwd": "/"},
+        "process": {"args": ["true"],
+                    "env": ["K3SAMD_VISIBLE_DEVICES=all"], "cwd": "/"},
         "root": {"path": "rootfs"},
         "linux": {},
     }

@@ -75,3 +75,24 @@ def test_allocate_to_injection_timesliced_pair(tmp_path):
                                 "/dev/dri/renderD129", "/dev/dri/card1"])
     finally:
         h.close()
+
+
+def test_no_gpu_limit_pod_gets_nothing(tmp_path):
+    """The scheduling-bypass hole closed: a pod that sets
+    runtimeClassName: amd but requests NO amd.com/gpu limit never passes
+    through Allocate, so its container has no allocation env/annotation —
+    the runtime must inject zero GPU devices (default-deny), keeping
+    kubelet's device accounting authoritative."""
+    h = PluginHarness(tmp_path, n_gpus=8, replicas=1, register=False)
+    try:
+        # no Allocate call at all — this pod never requested the resource
+        out = oci_transform(tmp_path, h.root, {"PATH": "/usr/bin"})
+        assert "/dev/kfd" not in json.dumps(out)
+        devs = out["linux"].get("devices", [])
+        assert devs == []
+        # the pre-existing default deny cgroup rule is untouched, and no
+        # allow rules were added
+        rules = out["linux"]["resources"]["devices"]
+        assert rules == [{"allow": False, "access": "rwm"}]
+    finally:
+        h.close()

@@ -36,18 +36,20 @@ BASE_SPEC = {
 }
 
 
-def transform(tmp_path, spec, n_gpus=8, extra_env=None):
+def transform(tmp_path, spec, n_gpus=8, extra_env=None, runtime_env=None):
     sysfs = build_tree(tmp_path / "sys", n_gpus=n_gpus)
     cfg = tmp_path / "config.json"
     spec = json.loads(json.dumps(spec))  # deep copy
     if extra_env:
         spec["process"]["env"].extend(extra_env)
     cfg.write_text(json.dumps(spec))
+    env = {"K3SAMD_SYSFS_ROOT": str(sysfs),
+           "K3SAMD_DEV_ROOT": str(tmp_path / "nonexistent-dev")}
+    if runtime_env:
+        env.update(runtime_env)
     proc = subprocess.run(
         [str(RUNTIME), "--transform-only", str(cfg)],
-        env={"K3SAMD_SYSFS_ROOT": str(sysfs),
-             "K3SAMD_DEV_ROOT": str(tmp_path / "nonexistent-dev")},
-        capture_output=True, text=True, timeout=60)
+        env=env, capture_output=True, text=True, timeout=60)
     assert proc.returncode == 0, proc.stderr
     return json.loads(cfg.read_text()), proc.stderr
 
@@ -56,8 +58,20 @@ def device_paths(spec):
     return [d["path"] for d in spec["linux"]["devices"]]
 
 
-def test_default_injects_all_gpus(tmp_path):
-    out, _ = transform(tmp_path, BASE_SPEC, n_gpus=8)
+def test_no_allocation_injects_nothing_by_default(tmp_path):
+    """Default-deny: runtimeClassName alone (no amd.com/gpu limit, so no
+    Allocate env/annotation) must grant ZERO GPUs — otherwise any pod
+    could bypass kubelet device accounting (the NVIDIA_VISIBLE_DEVICES=all
+    foot-gun the reference stack inherits, /root/reference/README.md:164)."""
+    out, err = transform(tmp_path, BASE_SPEC, n_gpus=8)
+    assert "/dev/kfd" not in json.dumps(out)
+    assert "devices" not in out["linux"] or not device_paths(out)
+    assert "no allocation" in err and "default-deny" in err
+
+
+def test_explicit_all_sentinel_injects_all_gpus(tmp_path):
+    out, _ = transform(tmp_path, BASE_SPEC, n_gpus=8,
+                       extra_env=["K3SAMD_VISIBLE_DEVICES=all"])
     paths = device_paths(out)
     assert "/dev/kfd" in paths
     for k in range(8):
@@ -71,6 +85,27 @@ def test_default_injects_all_gpus(tmp_path):
     assert len(allows) == 17
     for r in allows:
         assert r["type"] == "c" and r["access"] == "rwm"
+
+
+def test_allow_all_runtime_flag(tmp_path):
+    """Operator-level debug override: K3SAMD_ALLOW_ALL=1 on the runtime
+    binary (NOT the container env) restores inject-all for no-allocation
+    containers."""
+    out, _ = transform(tmp_path, BASE_SPEC, n_gpus=2,
+                       runtime_env={"K3SAMD_ALLOW_ALL": "1"})
+    paths = device_paths(out)
+    assert "/dev/kfd" in paths
+    assert len(paths) == 5  # kfd + 2x(render+card)
+
+
+def test_allow_all_not_honored_from_container_env(tmp_path):
+    """K3SAMD_ALLOW_ALL inside the container's own env must NOT grant all
+    GPUs — only the runtime binary's environment (operator-controlled)
+    may; otherwise any pod spec could set it."""
+    out, err = transform(tmp_path, BASE_SPEC, n_gpus=2,
+                         extra_env=["K3SAMD_ALLOW_ALL=1"])
+    assert "/dev/kfd" not in json.dumps(out)
+    assert "no allocation" in err
 
 
 def test_visible_devices_subset(tmp_path):
@@ -91,6 +126,7 @@ def test_visible_none_skips(tmp_path):
 
 def test_unrelated_fields_roundtrip(tmp_path):
     spec = json.loads(json.dumps(BASE_SPEC))
+    spec["process"]["env"].append("K3SAMD_VISIBLE_DEVICES=all")
     spec["annotations"] = {"io.kubernetes.pod.name": "rocm-smi"}
     spec["hooks"] = {"prestart": [{"path": "/bin/true"}]}
     spec["linux"]["seccomp"] = {"defaultAction": "SCMP_ACT_ALLOW"}
@@ -105,7 +141,9 @@ def test_unrelated_fields_roundtrip(tmp_path):
 def test_idempotent(tmp_path):
     sysfs = build_tree(tmp_path / "sys", n_gpus=2)
     cfg = tmp_path / "config.json"
-    cfg.write_text(json.dumps(BASE_SPEC))
+    spec = json.loads(json.dumps(BASE_SPEC))
+    spec["process"]["env"].append("K3SAMD_VISIBLE_DEVICES=all")
+    cfg.write_text(json.dumps(spec))
     env = {"K3SAMD_SYSFS_ROOT": str(sysfs),
            "K3SAMD_DEV_ROOT": str(tmp_path / "nodev")}
     for _ in range(2):
@@ -125,7 +163,8 @@ def test_rocm_mount_injection(tmp_path):
     sysfs = build_tree(tmp_path / "sys", n_gpus=1)
     cfg = tmp_path / "config.json"
     spec = json.loads(json.dumps(BASE_SPEC))
-    spec["process"]["env"].append("K3SAMD_INJECT_ROCM=1")
+    spec["process"]["env"] += ["K3SAMD_VISIBLE_DEVICES=all",
+                               "K3SAMD_INJECT_ROCM=1"]
     cfg.write_text(json.dumps(spec))
     subprocess.run(
         [str(RUNTIME), "--transform-only", str(cfg)],
@@ -148,7 +187,8 @@ def test_rocm_default_with_env_optout(tmp_path):
     sysfs = build_tree(tmp_path / "sys", n_gpus=1)
     cfg = tmp_path / "config.json"
     spec = json.loads(json.dumps(BASE_SPEC))
-    spec["process"]["env"].append("K3SAMD_INJECT_ROCM=0")
+    spec["process"]["env"] += ["K3SAMD_VISIBLE_DEVICES=all",
+                               "K3SAMD_INJECT_ROCM=0"]
     cfg.write_text(json.dumps(spec))
     subprocess.run(
         [str(RUNTIME), "--transform-only", str(cfg)],
@@ -174,7 +214,8 @@ def test_annotation_fallback_selection(tmp_path):
 
 
 def test_cpu_only_node_injects_nothing(tmp_path):
-    out, _ = transform(tmp_path, BASE_SPEC, n_gpus=0)
+    out, _ = transform(tmp_path, BASE_SPEC, n_gpus=0,
+                       extra_env=["K3SAMD_VISIBLE_DEVICES=all"])
     assert "/dev/kfd" not in json.dumps(out)
 
 
@@ -185,7 +226,10 @@ def test_runc_passthrough_create(tmp_path):
     sysfs = build_tree(tmp_path / "sys", n_gpus=1)
     bundle = tmp_path / "bundle"
     bundle.mkdir()
-    (bundle / "config.json").write_text(json.dumps(BASE_SPEC))
+    spec = json.loads(json.dumps(BASE_SPEC))
+    spec["process"]["env"].append(
+        "K3SAMD_VISIBLE_DEVICES=amdgpu-1a2b3c4d5e6f0000")
+    (bundle / "config.json").write_text(json.dumps(spec))
     # stub runc that records its argv
     stub = tmp_path / "runc-stub"
     argv_log = tmp_path / "argv.txt"
