@@ -1,6 +1,7 @@
 #include "metrics.h"
 
 #include <arpa/inet.h>
+#include <chrono>
 #include <cstring>
 #include <netinet/in.h>
 #include <sys/socket.h>
@@ -73,21 +74,36 @@ void MetricsServer::serve_loop() {
     if (lfd < 0) break;
     int cfd = ::accept(lfd, nullptr, nullptr);
     if (cfd < 0) break;
-    // read whatever request line arrives (we answer every request the same)
-    char buf[1024];
-    ssize_t ignored = ::read(cfd, buf, sizeof(buf));
-    (void)ignored;
-    std::string body = render_ ? render_() : "";
-    char head[160];
-    int n = std::snprintf(head, sizeof(head),
-                          "HTTP/1.0 200 OK\r\n"
-                          "Content-Type: text/plain; version=0.0.4\r\n"
-                          "Content-Length: %zu\r\n\r\n",
-                          body.size());
-    (void)::send(cfd, head, (size_t)n, MSG_NOSIGNAL);
-    (void)::send(cfd, body.data(), body.size(), MSG_NOSIGNAL);
-    ::close(cfd);
+    // A client that connects and sends nothing must not wedge the accept
+    // loop (the DaemonSet's livenessProbe hits this endpoint): bound both
+    // directions with short socket timeouts and serve each connection on
+    // its own thread so one slow peer never blocks the next probe.
+    timeval tv{2, 0};
+    ::setsockopt(cfd, SOL_SOCKET, SO_RCVTIMEO, &tv, sizeof(tv));
+    ::setsockopt(cfd, SOL_SOCKET, SO_SNDTIMEO, &tv, sizeof(tv));
+    active_.fetch_add(1);
+    std::thread([this, cfd] {
+      char buf[1024];
+      ssize_t ignored = ::read(cfd, buf, sizeof(buf));
+      (void)ignored;
+      std::string body;
+      if (!stopping_.load() && render_) body = render_();
+      char head[160];
+      int n = std::snprintf(head, sizeof(head),
+                            "HTTP/1.0 200 OK\r\n"
+                            "Content-Type: text/plain; version=0.0.4\r\n"
+                            "Content-Length: %zu\r\n\r\n",
+                            body.size());
+      (void)::send(cfd, head, (size_t)n, MSG_NOSIGNAL);
+      (void)::send(cfd, body.data(), body.size(), MSG_NOSIGNAL);
+      ::close(cfd);
+      active_.fetch_sub(1);
+    }).detach();
   }
+  // connection threads are bounded by the 2 s socket timeouts; wait for
+  // them so render_'s captures stay valid through stop()
+  for (int i = 0; i < 500 && active_.load() > 0; ++i)
+    std::this_thread::sleep_for(std::chrono::milliseconds(10));
 }
 
 }  // namespace k3samd

@@ -42,6 +42,7 @@ class MetricsServer {
   std::atomic<int> listen_fd_{-1};
   std::thread thread_;
   std::atomic<bool> stopping_{false};
+  std::atomic<int> active_{0};  // in-flight connection threads
 };
 
 }  // namespace k3samd

@@ -129,3 +129,44 @@ def test_metrics_endpoint(tmp_path):
     finally:
         proc.terminate()
         proc.wait(timeout=10)
+
+
+def test_silent_client_does_not_wedge(tmp_path):
+    """A client that connects and sends nothing must not block the
+    livenessProbe: while the silent connection is still open, a normal
+    scrape has to succeed promptly (each connection is served on its own
+    thread with a 2 s socket timeout)."""
+    import urllib.request
+    root = build_tree(tmp_path / "sys", n_gpus=1)
+    sock = str(tmp_path / "amd.sock")
+    port = free_port()
+    proc = subprocess.Popen(
+        [str(PLUGIN), "--plugin-sock", sock, "--no-register",
+         "--metrics-addr", f"127.0.0.1:{port}", "--health-poll-ms", "0"],
+        env={"K3SAMD_SYSFS_ROOT": str(root)},
+        stdout=subprocess.PIPE, stderr=subprocess.PIPE)
+    silent = []
+    try:
+        deadline = time.time() + 10
+        while time.time() < deadline:
+            try:
+                s = socket.create_connection(("127.0.0.1", port), timeout=2)
+                silent.append(s)  # connect, send nothing, keep it open
+                break
+            except OSError:
+                time.sleep(0.1)
+        assert silent, "endpoint never came up"
+        # two more silent clients for good measure
+        for _ in range(2):
+            silent.append(
+                socket.create_connection(("127.0.0.1", port), timeout=2))
+        t0 = time.time()
+        body = urllib.request.urlopen(
+            f"http://127.0.0.1:{port}/metrics", timeout=5).read().decode()
+        assert "k3samd_gpu_devices" in body
+        assert time.time() - t0 < 3, "scrape was blocked by silent clients"
+    finally:
+        for s in silent:
+            s.close()
+        proc.terminate()
+        proc.wait(timeout=10)
