@@ -178,19 +178,27 @@ struct H2Conn : std::enable_shared_from_this<H2Conn> {
   }
 
   // Send a DATA frame respecting connection + stream flow-control windows.
+  // Splits the payload to whatever window credit is currently available
+  // (a peer window smaller than the payload must produce several DATA
+  // frames, not a wait for credit that can never arrive).
   bool send_data(const std::shared_ptr<H2Stream>& st, std::string_view payload,
                  bool end_stream) {
+    if (payload.empty())
+      return send(frame_bytes(kData, end_stream ? kEndStream : 0, st->id,
+                              payload));
     size_t off = 0;
-    while (off < payload.size() || (payload.empty() && off == 0)) {
-      size_t chunk = std::min(payload.size() - off, kMaxFrame);
-      if (chunk > 0) {
+    while (off < payload.size()) {
+      size_t chunk;
+      {
         std::unique_lock<std::mutex> lk(mu);
         bool ok = window_cv.wait_for(lk, std::chrono::seconds(60), [&] {
           return closed.load() || st->cancelled.load() ||
-                 ((size_t)send_window.load() >= chunk &&
-                  (size_t)st->send_window.load() >= chunk);
+                 (send_window.load() > 0 && st->send_window.load() > 0);
         });
         if (!ok || closed.load() || st->cancelled.load()) return false;
+        chunk = std::min(
+            {payload.size() - off, kMaxFrame, (size_t)send_window.load(),
+             (size_t)st->send_window.load()});
         send_window.fetch_sub((int)chunk);
         st->send_window.fetch_sub((int)chunk);
       }
@@ -199,7 +207,6 @@ struct H2Conn : std::enable_shared_from_this<H2Conn> {
                             st->id, payload.substr(off, chunk))))
         return false;
       off += chunk;
-      if (payload.empty()) break;
     }
     return true;
   }
@@ -556,16 +563,29 @@ void GrpcServer::connection_loop(const std::shared_ptr<H2Conn>& conn) {
             first_grpc_message(st->data, req);
             auto uit = unary_.find(path);
             if (uit != unary_.end()) {
-              std::string resp;
-              GrpcStatus gs = uit->second(req, resp);
-              if (gs.code == 0) {
-                send_response_headers(conn, st);
-                conn->send_data(st, grpc_frame(resp), false);
-                send_trailers(conn, st, gs, false);
-              } else {
-                send_trailers(conn, st, gs, true);
-              }
-              finish_stream(st);
+              // Run on a worker thread, NOT this frame-reader thread: a
+              // response larger than the peer's flow-control window makes
+              // send_data block for WINDOW_UPDATE frames that only the
+              // frame-reader can deliver (reachable with big
+              // Allocate/GetPreferredAllocation payloads at high
+              // time-slicing replica counts).
+              UnaryHandler handler = uit->second;
+              spawn([conn, st, handler, finish_stream, req] {
+                std::string resp;
+                GrpcStatus gs = handler(req, resp);
+                if (st->cancelled.load() || conn->closed.load()) {
+                  finish_stream(st);
+                  return;
+                }
+                if (gs.code == 0) {
+                  send_response_headers(conn, st);
+                  conn->send_data(st, grpc_frame(resp), false);
+                  send_trailers(conn, st, gs, false);
+                } else {
+                  send_trailers(conn, st, gs, true);
+                }
+                finish_stream(st);
+              });
               continue;
             }
             auto sit = stream_.find(path);

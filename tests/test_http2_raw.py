@@ -178,3 +178,50 @@ def test_unknown_method_gets_unimplemented(plugin):
         assert status == 12  # UNIMPLEMENTED
     finally:
         conn.close()
+
+
+def test_unary_response_through_tiny_flow_control_window(plugin):
+    """A unary response larger than the peer's stream window must not
+    deadlock: the server's frame-reader has to keep consuming
+    WINDOW_UPDATE frames while the (worker-thread) unary handler waits
+    for window credit. Regression test for the reader-thread-sends-unary
+    design (grpc_transport.cc) — under the old inline dispatch this
+    stalled 60 s and failed the RPC."""
+    conn = RawConn(plugin)
+    try:
+        # shrink every stream's send window to 64 bytes
+        conn.send(frame(SETTINGS, 0, 0, struct.pack(">HI", 4, 64)))
+        ids = ["amdgpu-1a2b3c4d5e6f0000", "amdgpu-1a2b3c4d5e6f0001"]
+        conn.send(frame(HEADERS, END_HEADERS, 1,
+                        request_headers(b"/v1beta1.DevicePlugin/Allocate")))
+        conn.send(frame(DATA, END_STREAM, 1,
+                        grpc_frame(pb.encode_allocate_request([ids]))))
+        t0 = time.time()
+        resp_data = b""
+        done = False
+        while not done:
+            ftype, flags, stream, payload = conn.read_frame()
+            if ftype == SETTINGS and not flags & ACK:
+                conn.send(frame(SETTINGS, ACK, 0, b""))
+            elif ftype == DATA and stream == 1:
+                assert len(payload) <= 64, "server ignored the stream window"
+                resp_data += payload
+                # drip window credit back: stream + connection level
+                inc = struct.pack(">I", max(len(payload), 1))
+                conn.send(frame(WINDOW_UPDATE, 0, 1, inc))
+                conn.send(frame(WINDOW_UPDATE, 0, 0, inc))
+            elif ftype == HEADERS and stream == 1 and \
+                    b"grpc-status" in payload:
+                idx = payload.find(b"grpc-status")
+                vlen = payload[idx + 11]
+                assert int(payload[idx + 12: idx + 12 + vlen]) == 0
+                done = True
+        assert time.time() - t0 < 20, "flow-control deadlock"
+        assert resp_data[:1] == b"\x00"
+        mlen = int.from_bytes(resp_data[1:5], "big")
+        assert len(resp_data) >= 5 + mlen > 64  # actually exercised chunking
+        resp = pb.decode_allocate_response(resp_data[5:5 + mlen])
+        env = resp[0]["envs"]
+        assert env["K3SAMD_VISIBLE_DEVICES"] == ",".join(ids)
+    finally:
+        conn.close()
