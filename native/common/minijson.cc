@@ -71,14 +71,42 @@ struct Parser {
               else if (h >= 'A' && h <= 'F') cp |= (unsigned)(h - 'A' + 10);
               else fail("bad \\u digit");
             }
-            // encode as UTF-8 (surrogate pairs handled as two escapes)
+            // A high surrogate must combine with the following \uDC00..
+            // \uDFFF escape into one code point — encoding the halves
+            // separately produces CESU-8, which runc's JSON parser can
+            // reject (non-BMP chars, e.g. emoji in env values).
+            if (cp >= 0xD800 && cp <= 0xDBFF) {
+              if (end - p >= 6 && p[0] == '\\' && p[1] == 'u') {
+                unsigned lo = 0;
+                bool ok = true;
+                for (int i = 2; i < 6; ++i) {
+                  char h = p[i];
+                  lo <<= 4;
+                  if (h >= '0' && h <= '9') lo |= (unsigned)(h - '0');
+                  else if (h >= 'a' && h <= 'f') lo |= (unsigned)(h - 'a' + 10);
+                  else if (h >= 'A' && h <= 'F') lo |= (unsigned)(h - 'A' + 10);
+                  else { ok = false; break; }
+                }
+                if (ok && lo >= 0xDC00 && lo <= 0xDFFF) {
+                  p += 6;
+                  cp = 0x10000 + ((cp - 0xD800) << 10) + (lo - 0xDC00);
+                }
+              }
+              // unpaired surrogate falls through and is encoded as-is
+              // (matches lenient JSON parsers; round-trips our output)
+            }
             if (cp < 0x80) {
               out += (char)cp;
             } else if (cp < 0x800) {
               out += (char)(0xC0 | (cp >> 6));
               out += (char)(0x80 | (cp & 0x3F));
-            } else {
+            } else if (cp < 0x10000) {
               out += (char)(0xE0 | (cp >> 12));
+              out += (char)(0x80 | ((cp >> 6) & 0x3F));
+              out += (char)(0x80 | (cp & 0x3F));
+            } else {
+              out += (char)(0xF0 | (cp >> 18));
+              out += (char)(0x80 | ((cp >> 12) & 0x3F));
               out += (char)(0x80 | ((cp >> 6) & 0x3F));
               out += (char)(0x80 | (cp & 0x3F));
             }

@@ -256,6 +256,15 @@ int test_minijson() {
   // \u escape
   auto u = k3samd::json_parse("\"\\u00e9\"");
   CHECK(u->str == "\xc3\xa9");
+  // surrogate pair combines into ONE 4-byte UTF-8 code point (U+1F600),
+  // not two 3-byte CESU-8 halves runc's parser would reject
+  auto emoji = k3samd::json_parse("\"\\ud83d\\ude00\"");
+  CHECK(emoji->str == "\xf0\x9f\x98\x80");
+  // and the raw UTF-8 round-trips through serialize -> parse untouched
+  CHECK(k3samd::json_parse(k3samd::json_serialize(emoji))->str ==
+        "\xf0\x9f\x98\x80");
+  // unpaired high surrogate stays lenient (no throw)
+  CHECK(!k3samd::json_parse("\"\\ud83d x\"")->str.empty());
   bool threw = false;
   try {
     k3samd::json_parse("{\"unterminated\": ");

@@ -213,6 +213,29 @@ def test_annotation_fallback_selection(tmp_path):
                                  "/dev/dri/card1"]
 
 
+def test_non_bmp_env_value_roundtrips_valid_utf8(tmp_path):
+    """Emoji (non-BMP) in env values/annotations must come back as real
+    UTF-8, whether escaped (\\ud83d\\ude00 surrogate pair -> one 4-byte
+    code point) or raw — CESU-8 output would make runc reject the spec."""
+    spec = json.loads(json.dumps(BASE_SPEC))
+    spec["process"]["env"].append("GREETING=hi \N{GRINNING FACE}")
+    spec["annotations"] = {"note": "party \N{PARTY POPPER}"}
+    sysfs = build_tree(tmp_path / "sys", n_gpus=1)
+    cfg = tmp_path / "config.json"
+    # write with escaped surrogates (ensure_ascii=True is json.dumps default)
+    cfg.write_text(json.dumps(spec, ensure_ascii=True))
+    subprocess.run(
+        [str(RUNTIME), "--transform-only", str(cfg)],
+        env={"K3SAMD_SYSFS_ROOT": str(sysfs),
+             "K3SAMD_DEV_ROOT": str(tmp_path / "nodev")},
+        check=True, capture_output=True, timeout=60)
+    raw = cfg.read_bytes()
+    raw.decode("utf-8")  # must be valid UTF-8 (CESU-8 halves are not)
+    out = json.loads(raw)
+    assert "GREETING=hi \N{GRINNING FACE}" in out["process"]["env"]
+    assert out["annotations"]["note"] == "party \N{PARTY POPPER}"
+
+
 def test_cpu_only_node_injects_nothing(tmp_path):
     out, _ = transform(tmp_path, BASE_SPEC, n_gpus=0,
                        extra_env=["K3SAMD_VISIBLE_DEVICES=all"])
