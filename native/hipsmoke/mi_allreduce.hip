@@ -105,23 +105,37 @@ int main(int argc, char** argv) {
     };
     run_once();  // warm
     for (int i = 0; i < ngpus; ++i) HIP_CHECK(hipStreamSynchronize(streams[i]));
-    hipEvent_t ev0, ev1;
-    HIP_CHECK(hipSetDevice(0));
-    HIP_CHECK(hipEventCreate(&ev0));
-    HIP_CHECK(hipEventCreate(&ev1));
-    HIP_CHECK(hipEventRecord(ev0, streams[0]));
+    // Time with per-device events and take the MAX elapsed across devices:
+    // a rank-0-only window under-reports when other ranks' streams drain
+    // later (skewed launch order inflates busbw at N>1).
+    std::vector<hipEvent_t> ev0(ngpus), ev1(ngpus);
+    for (int i = 0; i < ngpus; ++i) {
+      HIP_CHECK(hipSetDevice(i));
+      HIP_CHECK(hipEventCreate(&ev0[i]));
+      HIP_CHECK(hipEventCreate(&ev1[i]));
+      HIP_CHECK(hipEventRecord(ev0[i], streams[i]));
+    }
     for (int it = 0; it < iters; ++it) run_once();
-    HIP_CHECK(hipEventRecord(ev1, streams[0]));
+    for (int i = 0; i < ngpus; ++i) {
+      HIP_CHECK(hipSetDevice(i));
+      HIP_CHECK(hipEventRecord(ev1[i], streams[i]));
+    }
     for (int i = 0; i < ngpus; ++i) HIP_CHECK(hipStreamSynchronize(streams[i]));
-    float ms;
-    HIP_CHECK(hipEventElapsedTime(&ms, ev0, ev1));
+    float ms = 0;
+    for (int i = 0; i < ngpus; ++i) {
+      float mi;
+      HIP_CHECK(hipEventElapsedTime(&mi, ev0[i], ev1[i]));
+      if (mi > ms) ms = mi;
+    }
     double t = ms / 1e3 / iters;
     double algbw = bytes / t / 1e9;
     double busbw = algbw * 2.0 * (ngpus - 1) / ngpus;
     last_busbw = busbw;
     std::printf("| %9zu | %10.1f | %10.1f |\n", bytes, algbw, busbw);
-    HIP_CHECK(hipEventDestroy(ev0));
-    HIP_CHECK(hipEventDestroy(ev1));
+    for (int i = 0; i < ngpus; ++i) {
+      HIP_CHECK(hipEventDestroy(ev0[i]));
+      HIP_CHECK(hipEventDestroy(ev1[i]));
+    }
   }
   std::printf("+-----------+------------+------------+\n");
   // xGMI sanity verdict: on the fully-connected MI355X node
