@@ -48,6 +48,15 @@ def parse_args(argv=None):
     return p.parse_args(argv)
 
 
+# multi-GPU self-diagnosis lives in the package so the in-pod payloads
+# (k3samd.parallel.allreduce) share the exact same logic
+from k3samd.utils.rccl_diag import (  # noqa: E402
+    assert_unique_device_binding, canonical_device_ids,
+    enable_rccl_debug_capture, gather_transport_counts,
+    parse_rccl_transports,
+)
+
+
 def build_result(*, value, n_gpus, steps, warmup, ms_per_step, buffer_mib,
                  variant):
     """Assemble the contract JSON (separated out for CPU-side tests)."""
@@ -88,30 +97,51 @@ def main(argv=None) -> int:
         print(f"# note: --gpus {args.gpus} != WORLD_SIZE {world_size}; "
               f"reporting n_gpus={world_size}", file=sys.stderr)
 
-    if not torch.cuda.is_available():
-        print("bench.py requires a GPU (run on the MI355X box)",
+    # CPU rehearsal of the exact torchrun path (gloo, no GPU): same
+    # rendezvous, barriers, max-over-ranks reduction and JSON contract,
+    # with a torch-eager triad standing in for the HIP kernel. The output
+    # is stamped "rehearsal" so it can never pass as a measurement.
+    rehearsal = os.environ.get("K3SAMD_BENCH_REHEARSAL") == "1"
+
+    if not rehearsal and not torch.cuda.is_available():
+        print("bench.py requires a GPU (run on the MI355X box; "
+              "set K3SAMD_BENCH_REHEARSAL=1 for the CPU dry-run)",
               file=sys.stderr)
         return 1
 
+    rccl_log = None
     dist = None
     if world_size > 1:
         import torch.distributed as dist_mod
         dist = dist_mod
+        if not rehearsal:
+            rccl_log = enable_rccl_debug_capture()
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "29571")
-        dist.init_process_group(backend="nccl", rank=rank,
-                                world_size=world_size)
+        dist.init_process_group(backend="gloo" if rehearsal else "nccl",
+                                rank=rank, world_size=world_size)
 
-    torch.cuda.set_device(local_rank)
-    device = torch.device("cuda", local_rank)
+    if rehearsal:
+        device = torch.device("cpu")
+    else:
+        torch.cuda.set_device(local_rank)
+        device = torch.device("cuda", local_rank)
+        if dist is not None:
+            # fail loudly if two ranks share a physical GPU (halves the
+            # aggregate silently otherwise)
+            assert_unique_device_binding(dist, device, rank)
 
-    from k3samd import ops
-    if not ops.native_available():
-        print("k3samd native extension missing — refusing to run a fallback",
-              file=sys.stderr)
-        return 2
+    if not rehearsal:
+        from k3samd import ops
+        if not ops.native_available():
+            print("k3samd native extension missing — refusing to run a "
+                  "fallback", file=sys.stderr)
+            return 2
 
-    n = args.buffer_mib * (1 << 20) // 4  # fp32 elements per buffer
+    buffer_mib = args.buffer_mib
+    if rehearsal:
+        buffer_mib = min(buffer_mib, 4)  # keep the dry-run light
+    n = buffer_mib * (1 << 20) // 4  # fp32 elements per buffer
     b = torch.rand(n, device=device)
     c = torch.rand(n, device=device)
     a = torch.empty_like(b)
@@ -119,23 +149,30 @@ def main(argv=None) -> int:
     step_bytes = 3 * buffer_bytes
 
     def run_step(nt: bool):
-        ops.stream_triad(a, b, c, args.scalar, nontemporal=nt)
+        if rehearsal:
+            torch.add(b, c, alpha=args.scalar, out=a)
+        else:
+            ops.stream_triad(a, b, c, args.scalar, nontemporal=nt)
+
+    def sync():
+        if not rehearsal:
+            torch.cuda.synchronize()
 
     def barrier():
         if dist is not None:
             dist.barrier()
 
     # ---- variant selection (globally consistent across ranks) ----
-    if args.variant == "auto":
+    if args.variant == "auto" and not rehearsal:
         timings = []
         for nt in (False, True):
             for _ in range(5):
                 run_step(nt)
-            torch.cuda.synchronize()
+            sync()
             t0 = time.perf_counter()
             for _ in range(10):
                 run_step(nt)
-            torch.cuda.synchronize()
+            sync()
             timings.append(time.perf_counter() - t0)
         # nccl collectives need device tensors
         t = torch.tensor(timings, dtype=torch.float64, device=device)
@@ -148,7 +185,7 @@ def main(argv=None) -> int:
 
     # ---- optional hipGraph capture of one step ----
     graph = None
-    if args.graph in ("auto", "on"):
+    if args.graph in ("auto", "on") and not rehearsal:
         try:
             g = torch.cuda.CUDAGraph()
             # side-stream warmup required before capture
@@ -176,15 +213,15 @@ def main(argv=None) -> int:
     # ---- warmup ----
     for _ in range(args.warmup):
         do_step()
-    torch.cuda.synchronize()
+    sync()
 
     # ---- timed region ----
     barrier()
-    torch.cuda.synchronize()
+    sync()
     t0 = time.perf_counter()
     for _ in range(args.steps):
         do_step()
-    torch.cuda.synchronize()
+    sync()
     elapsed = time.perf_counter() - t0
     barrier()
 
@@ -196,11 +233,27 @@ def main(argv=None) -> int:
     value = n_gpus * step_bytes * args.steps / t_max / 1e9
     ms_per_step = t_max / args.steps * 1e3
 
+    # ---- RCCL transport self-diagnosis (N>1 on the GPU path) ----
+    transports = None
+    if dist is not None and rccl_log is not None:
+        transports = gather_transport_counts(dist, rccl_log, rank, world_size)
+        if rank == 0:
+            print(f"# rccl transports: {transports or 'no log captured'}",
+                  file=sys.stderr)
+            non_p2p = sum(v for k, v in transports.items() if k != "P2P")
+            if transports and (transports.get("P2P", 0) == 0 or non_p2p):
+                print("# WARNING: RCCL channels not all xGMI P2P — check "
+                      "device injection / IPC mode", file=sys.stderr)
+
     if rank == 0:
         res = build_result(value=value, n_gpus=n_gpus, steps=args.steps,
                            warmup=args.warmup, ms_per_step=ms_per_step,
-                           buffer_mib=args.buffer_mib, variant=variant)
+                           buffer_mib=buffer_mib, variant=variant)
         res["config"]["hip_graph"] = graph is not None
+        if rehearsal:
+            res["rehearsal"] = True  # gloo/CPU dry-run, NOT a measurement
+        if transports is not None:
+            res["config"]["rccl_transports"] = transports
         print(json.dumps(res))
 
     if dist is not None:

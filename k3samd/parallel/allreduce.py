@@ -97,17 +97,27 @@ def main(argv=None) -> int:
     backend = args.backend
     if backend == "auto":
         backend = "nccl" if torch.cuda.is_available() else "gloo"
-    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
-    os.environ.setdefault("MASTER_PORT", "29572")
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    rccl_log = None
+    if backend == "nccl" and world > 1:
+        # capture RCCL's channel-setup log so the run can prove (or
+        # refute) xGMI p2p transport selection on its own
+        from k3samd.utils.rccl_diag import enable_rccl_debug_capture
+        rccl_log = enable_rccl_debug_capture()
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29572")
     dist.init_process_group(backend=backend, rank=rank, world_size=world)
 
     device = None
     if backend == "nccl":
         torch.cuda.set_device(local_rank)
         device = torch.device("cuda", local_rank)
+        if world > 1:
+            # two ranks on one physical GPU would corrupt the busbw claim
+            from k3samd.utils.rccl_diag import assert_unique_device_binding
+            assert_unique_device_binding(dist, device, rank)
 
     sizes = []
     b = args.min_mib << 20
@@ -116,13 +126,31 @@ def main(argv=None) -> int:
         b *= 4
     rows = busbw_sweep(sizes, iters=args.iters, device=device, op=args.op)
 
+    transports = None
+    if rccl_log is not None:
+        from k3samd.utils.rccl_diag import gather_transport_counts
+        transports = gather_transport_counts(dist, rccl_log, rank, world)
+
     if rank == 0:
         for r in rows:
             print(f"{r['bytes']:>12d} B  algbw {r['algbw_gbps']:8.2f} GB/s"
                   f"  busbw {r['busbw_gbps']:8.2f} GB/s")
-        print(json.dumps({"payload": "allreduce", "backend": backend,
-                          "op": args.op, "world_size": world,
-                          "rows": rows}))
+        result = {"payload": "allreduce", "backend": backend,
+                  "op": args.op, "world_size": world, "rows": rows}
+        if backend == "nccl" and world > 1:
+            # xGMI sanity verdict: the fully-connected
+            # MI355X node must beat one xGMI link's ~153 GB/s at large
+            # sizes, and every RCCL channel should be P2P (SHM = host
+            # bounce = the injected /dev/dri set is broken)
+            best = max(r["busbw_gbps"] for r in rows)
+            result["max_busbw_gbps"] = best
+            result["xgmi_p2p_ok"] = best > 153.0
+            if transports is not None:
+                result["rccl_transports"] = transports
+                non_p2p = sum(v for k, v in transports.items() if k != "P2P")
+                result["rccl_all_p2p"] = bool(transports.get("P2P")) \
+                    and non_p2p == 0
+        print(json.dumps(result))
     dist.destroy_process_group()
     return 0
 

@@ -91,3 +91,24 @@ def test_cli_single_rank():
     last = proc.stdout.strip().splitlines()[-1]
     j = json.loads(last)
     assert j["backend"] == "gloo" and j["world_size"] == 1
+
+
+def test_torchrun_world4_cli(tmp_path):
+    """The exact in-pod launch form (torch.distributed.run, the module
+    CLI) rehearsed at width 4 on gloo — the same code path the 8-GPU
+    RCCL pod runs, minus the backend."""
+    proc = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run",
+         "--nnodes=1", "--nproc-per-node=4",
+         "--master-addr", "127.0.0.1", "--master-port", free_port(),
+         "-m", "k3samd.parallel.allreduce",
+         "--backend", "gloo", "--min-mib", "1", "--max-mib", "1",
+         "--iters", "2"],
+        cwd=str(REPO), env=dict(os.environ), capture_output=True, text=True,
+        timeout=600)
+    assert proc.returncode == 0, proc.stderr[-3000:]
+    last = [ln for ln in proc.stdout.splitlines() if ln.startswith("{")]
+    assert len(last) == 1
+    j = json.loads(last[0])
+    assert j["world_size"] == 4
+    assert j["rows"][0]["busbw_gbps"] > 0
