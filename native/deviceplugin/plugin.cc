@@ -4,6 +4,7 @@
 #include <chrono>
 #include <cstdio>
 #include <fstream>
+#include <map>
 #include <set>
 #include <sstream>
 #include <sys/stat.h>
@@ -30,6 +31,18 @@ bool PluginConfig::from_yaml(const std::string& text, PluginConfig& out,
       // mirror the reference's `migStrategy: none` (values.yaml:11) only.
       if (err) *err = "only migStrategy 'none' is supported";
       return false;
+    }
+    // additive `health:` block — RAS/ECC thresholds (gpu_health.h). A
+    // value of -1 disables that check.
+    if (const YNode* h = root.get("health")) {
+      if (const YNode* n = h->get("maxUncorrectableErrors"))
+        out.health.max_uncorrectable = n->as_int(0);
+      if (const YNode* n = h->get("maxCorrectableErrors"))
+        out.health.max_correctable = n->as_int(10000);
+      if (const YNode* n = h->get("maxPcieReplays"))
+        out.health.max_pcie_replays = n->as_int(-1);
+      if (const YNode* n = h->get("maxResets"))
+        out.health.max_resets = n->as_int(0);
     }
     const YNode* ts = root.get_path("sharing.timeSlicing");
     if (ts) {
@@ -287,15 +300,39 @@ bool DevicePlugin::poll_health_once() {
   std::set<std::string> present;
   for (const auto& g : topo.gpus) present.insert(g.stable_id());
 
+  // RAS/error-state probe outside the lock (sysfs reads). A GPU is
+  // unhealthy when it VANISHED from KFD *or* its error counters trip the
+  // configured thresholds (sick-but-present: HBM ECC UEs, reset events).
+  std::map<int, std::string> sick;  // gpu_index -> reason
+  {
+    std::lock_guard<std::recursive_mutex> lk(mu_);
+    for (size_t gi = 0; gi < topo_.gpus.size(); ++gi) {
+      GpuHealthCounters c =
+          read_gpu_health(sysfs_root_, topo_.gpus[gi].card_index);
+      std::string reason = health_verdict(c, cfg_.health);
+      if (!reason.empty()) sick[(int)gi] = std::move(reason);
+    }
+  }
+
   std::lock_guard<std::recursive_mutex> lk(mu_);
   bool changed = false;
   for (auto& vd : devices_) {
     const auto& g = topo_.gpus[vd.gpu_index];
     bool healthy = present.count(g.stable_id()) > 0;
+    auto it = sick.find(vd.gpu_index);
+    if (healthy && it != sick.end()) healthy = false;
     if (healthy != vd.healthy) {
       vd.healthy = healthy;
       metrics_.health_transitions_total.fetch_add(1);
       changed = true;
+      if (!healthy)
+        std::fprintf(stderr, "k3samd-device-plugin: %s -> Unhealthy (%s)\n",
+                     vd.id.c_str(),
+                     it != sick.end() ? it->second.c_str()
+                                      : "KFD node disappeared");
+      else
+        std::fprintf(stderr, "k3samd-device-plugin: %s -> Healthy\n",
+                     vd.id.c_str());
     }
   }
   if (changed) {

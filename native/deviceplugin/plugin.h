@@ -15,6 +15,7 @@
 #include <vector>
 
 #include "../grpc/grpc_transport.h"
+#include "../topology/gpu_health.h"
 #include "../topology/kfd_topology.h"
 #include "dp_messages.h"
 #include "metrics.h"
@@ -31,6 +32,8 @@ struct PluginConfig {
   std::string resource_name = "amd.com/gpu";
   bool use_cdi = false;                // Allocate returns CDI device names
   std::string cdi_kind = "amd.com/gpu";
+  HealthPolicy health;                 // additive `health:` block (RAS/ECC
+                                       // thresholds; see gpu_health.h)
 
   // Parse the `version: v1 / flags / sharing.timeSlicing` YAML document.
   static bool from_yaml(const std::string& text, PluginConfig& out,

@@ -119,3 +119,20 @@ def build_tree(
             (hw / "power1_average").write_text("135000000\n")  # 135 W
 
     return root
+
+
+def make_gpu_sick(root, card_index, *, umc_ue=0, umc_ce=0, gfx_ue=0,
+                  gfx_ce=0, pcie_replays=None, resets=None):
+    """Write amdgpu RAS/error-state files for a present-but-sick GPU
+    (native/topology/gpu_health.cc reads these): per-block
+    `ras/<block>_err_count` ("ue: N\\nce: N"), `pcie_replay_count`,
+    `reset_count`. Call again with zeros to 'heal' the GPU."""
+    dev = Path(root) / f"class/drm/card{card_index}/device"
+    ras = dev / "ras"
+    ras.mkdir(parents=True, exist_ok=True)
+    (ras / "umc_err_count").write_text(f"ue: {umc_ue}\nce: {umc_ce}\n")
+    (ras / "gfx_err_count").write_text(f"ue: {gfx_ue}\nce: {gfx_ce}\n")
+    if pcie_replays is not None:
+        (dev / "pcie_replay_count").write_text(f"{pcie_replays}\n")
+    if resets is not None:
+        (dev / "reset_count").write_text(f"{resets}\n")

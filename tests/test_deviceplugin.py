@@ -296,6 +296,61 @@ def test_health_transition_pushes_update(harness):
     assert by_health == ["Healthy", "Unhealthy"]
 
 
+def test_sick_gpu_ras_ue_marked_unhealthy(harness):
+    """Sick-but-PRESENT GPU: an uncorrectable
+    RAS error (HBM ECC UE) must flip the device to Unhealthy over the
+    live ListAndWatch even though its KFD node still exists — kubelet
+    then stops scheduling pods onto it."""
+    from sysfs_builder import make_gpu_sick
+    h = harness(n_gpus=2, replicas=1, register=False, health_poll_ms=100)
+    stream = h.stream("ListAndWatch", timeout=30)
+    first = pb.decode_list_and_watch(next(stream))
+    assert all(d["health"] == "Healthy" for d in first)
+    make_gpu_sick(h.root, card_index=1, umc_ue=1)
+    second = pb.decode_list_and_watch(next(stream))
+    health = {d["id"]: d["health"] for d in second}
+    assert sorted(health.values()) == ["Healthy", "Unhealthy"]
+    # the SICK gpu (card 1 = unique_id ...0001) is the unhealthy one
+    sick_id = [i for i, hl in health.items() if hl == "Unhealthy"][0]
+    assert sick_id.endswith("0001")
+    # healing (RAS counters reset, e.g. after page retirement + reboot)
+    make_gpu_sick(h.root, card_index=1, umc_ue=0)
+    third = pb.decode_list_and_watch(next(stream))
+    assert all(d["health"] == "Healthy" for d in third)
+
+
+def test_sick_gpu_reset_event_marked_unhealthy(harness):
+    """A completed GPU reset (recovered hang) should drain the device."""
+    from sysfs_builder import make_gpu_sick
+    h = harness(n_gpus=1, replicas=2, register=False, health_poll_ms=100)
+    stream = h.stream("ListAndWatch", timeout=30)
+    first = pb.decode_list_and_watch(next(stream))
+    assert len(first) == 2
+    make_gpu_sick(h.root, card_index=0, resets=1)
+    second = pb.decode_list_and_watch(next(stream))
+    # BOTH time-slice replicas of the reset GPU go Unhealthy
+    assert [d["health"] for d in second] == ["Unhealthy", "Unhealthy"]
+
+
+def test_health_thresholds_configurable(harness):
+    """Correctable-error budget comes from the additive `health:` config
+    block; under the threshold stays Healthy, over it flips."""
+    from sysfs_builder import make_gpu_sick
+    cfg = DEFAULT_CFG.format(replicas=1) + (
+        "health:\n"
+        "  maxCorrectableErrors: 100\n"
+        "  maxResets: -1\n")
+    h = harness(n_gpus=1, replicas=1, register=False, health_poll_ms=100,
+                cfg_text=cfg)
+    stream = h.stream("ListAndWatch", timeout=30)
+    assert pb.decode_list_and_watch(next(stream))[0]["health"] == "Healthy"
+    make_gpu_sick(h.root, card_index=0, umc_ce=50, gfx_ce=30)  # 80 <= 100
+    time.sleep(0.5)  # two poll periods; no update should be pushed
+    make_gpu_sick(h.root, card_index=0, umc_ce=90, gfx_ce=30)  # 120 > 100
+    second = pb.decode_list_and_watch(next(stream))
+    assert second[0]["health"] == "Unhealthy"
+
+
 def test_config_hot_reload(harness, tmp_path):
     """Editing the mounted config (kubelet updates ConfigMap mounts in
     place) re-fans-out the device list and pushes it over the live
