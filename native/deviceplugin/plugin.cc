@@ -382,7 +382,10 @@ std::string DevicePlugin::render_metrics() {
       "# TYPE k3samd_gpu_busy_percent gauge\n"
       "# TYPE k3samd_gpu_temp_celsius gauge\n"
       "# TYPE k3samd_gpu_power_watts gauge\n"
-      "# TYPE k3samd_gpu_vram_used_bytes gauge\n";
+      "# TYPE k3samd_gpu_vram_used_bytes gauge\n"
+      "# TYPE k3samd_gpu_ras_uncorrectable_errors gauge\n"
+      "# TYPE k3samd_gpu_ras_correctable_errors gauge\n"
+      "# TYPE k3samd_gpu_reset_count gauge\n";
   for (const auto& g : gpus) {
     GpuRuntimeStats st = read_runtime_stats(sysfs_root_, g.card_index);
     char line[512];
@@ -397,6 +400,17 @@ std::string DevicePlugin::render_metrics() {
                   g.stable_id().c_str(),
                   st.power_uw < 0 ? -1.0 : st.power_uw / 1e6,
                   g.stable_id().c_str(), (unsigned long long)st.vram_used);
+    per_gpu += line;
+    // RAS/error-state gauges (-1 = not exposed by this driver build) —
+    // the same counters the health model thresholds on, so an operator
+    // can alert BEFORE a GPU flips Unhealthy
+    GpuHealthCounters hc = read_gpu_health(sysfs_root_, g.card_index);
+    std::snprintf(line, sizeof(line),
+                  "k3samd_gpu_ras_uncorrectable_errors{gpu=\"%s\"} %ld\n"
+                  "k3samd_gpu_ras_correctable_errors{gpu=\"%s\"} %ld\n"
+                  "k3samd_gpu_reset_count{gpu=\"%s\"} %ld\n",
+                  g.stable_id().c_str(), hc.ras_ue, g.stable_id().c_str(),
+                  hc.ras_ce, g.stable_id().c_str(), hc.reset_count);
     per_gpu += line;
   }
   char buf[2048];

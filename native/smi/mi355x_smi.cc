@@ -20,6 +20,7 @@
 
 #include "../common/json_writer.h"
 #include "../common/version.h"
+#include "../topology/gpu_health.h"
 #include "../topology/kfd_topology.h"
 
 namespace {
@@ -92,6 +93,15 @@ static int print_report(bool json) {
       w.key("busy_percent").value((int64_t)st.busy_percent);
       w.key("temp_milli_c").value((int64_t)st.temp_mc);
       w.key("power_uw").value((int64_t)st.power_uw);
+      // RAS/error state (same probe the device plugin's health model
+      // uses, so the operator sees exactly what kubelet is told)
+      k3samd::GpuHealthCounters hc =
+          k3samd::read_gpu_health(root, g.card_index);
+      w.key("ras_supported").value(hc.ras_present ? "true" : "false");
+      w.key("ras_uncorrectable").value((int64_t)hc.ras_ue);
+      w.key("ras_correctable").value((int64_t)hc.ras_ce);
+      w.key("pcie_replay_count").value((int64_t)hc.pcie_replay);
+      w.key("reset_count").value((int64_t)hc.reset_count);
       w.end_obj();
     }
     w.end_arr();
@@ -103,9 +113,9 @@ static int print_report(bool json) {
   std::printf("+---------------------------------------------------------------------------+\n");
   std::printf("| mi355x-smi            driver: %-12s                 k3samd stack   |\n",
               topo.driver_version.empty() ? "unknown" : topo.driver_version.c_str());
-  std::printf("+----+----------------------+--------+-----+-------------------+------+-----+------+\n");
-  std::printf("| ## | Name                 | Arch   | CUs | VRAM used / total | xGMI | Tmp | Pwr  |\n");
-  std::printf("+----+----------------------+--------+-----+-------------------+------+-----+------+\n");
+  std::printf("+----+----------------------+--------+-----+-------------------+------+-----+------+-----------+\n");
+  std::printf("| ## | Name                 | Arch   | CUs | VRAM used / total | xGMI | Tmp | Pwr  | ECC ue/ce |\n");
+  std::printf("+----+----------------------+--------+-----+-------------------+------+-----+------+-----------+\n");
   int i = 0;
   for (const auto& g : topo.gpus) {
     CardStats st = card_stats(root, g.card_index);
@@ -122,14 +132,21 @@ static int print_report(bool json) {
       std::snprintf(pwr, sizeof(pwr), "%4ldW", st.power_uw / 1000000);
     else
       std::snprintf(pwr, sizeof(pwr), "  - ");
-    std::printf("| %2d | %-20.20s | %-6s | %3u | %-17s | %4d | %s | %s |\n",
+    k3samd::GpuHealthCounters hc =
+        k3samd::read_gpu_health(root, g.card_index);
+    char ecc[16];
+    if (hc.ras_present)
+      std::snprintf(ecc, sizeof(ecc), "%ld/%ld", hc.ras_ue, hc.ras_ce);
+    else
+      std::snprintf(ecc, sizeof(ecc), "n/a");
+    std::printf("| %2d | %-20.20s | %-6s | %3u | %-17s | %4d | %s | %s | %-9s |\n",
                 i++, g.name.c_str(), g.gfx_arch().c_str(), g.compute_units(),
-                vram, g.xgmi_links, temp, pwr);
+                vram, g.xgmi_links, temp, pwr, ecc);
   }
   if (topo.gpus.empty()) {
     std::printf("| no AMD GPUs found (no KFD topology under %s)\n",
                 root.c_str());
   }
-  std::printf("+----+----------------------+--------+-----+-------------------+------+-----+------+\n");
+  std::printf("+----+----------------------+--------+-----+-------------------+------+-----+------+-----------+\n");
   return 0;
 }

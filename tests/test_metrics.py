@@ -170,3 +170,39 @@ def test_silent_client_does_not_wedge(tmp_path):
             s.close()
         proc.terminate()
         proc.wait(timeout=10)
+
+
+def test_ras_gauges_exposed(tmp_path):
+    """The health model's RAS counters appear as per-GPU gauges so an
+    operator can alert before a GPU flips Unhealthy."""
+    import urllib.request
+    from sysfs_builder import make_gpu_sick
+    root = build_tree(tmp_path / "sys", n_gpus=1)
+    make_gpu_sick(root, 0, umc_ue=2, umc_ce=41, resets=1)
+    sock = str(tmp_path / "amd.sock")
+    port = free_port()
+    proc = subprocess.Popen(
+        [str(PLUGIN), "--plugin-sock", sock, "--no-register",
+         "--metrics-addr", f"127.0.0.1:{port}", "--health-poll-ms", "0"],
+        env={"K3SAMD_SYSFS_ROOT": str(root)},
+        stdout=subprocess.PIPE, stderr=subprocess.PIPE)
+    try:
+        deadline = time.time() + 10
+        body = None
+        while time.time() < deadline and proc.poll() is None:
+            try:
+                body = urllib.request.urlopen(
+                    f"http://127.0.0.1:{port}/metrics", timeout=5
+                ).read().decode()
+                break
+            except OSError:
+                time.sleep(0.1)
+        assert body
+        m = parse_metrics(body)
+        gid = "amdgpu-1a2b3c4d5e6f0000"
+        assert m[f'k3samd_gpu_ras_uncorrectable_errors{{gpu="{gid}"}}'] == 2
+        assert m[f'k3samd_gpu_ras_correctable_errors{{gpu="{gid}"}}'] == 41
+        assert m[f'k3samd_gpu_reset_count{{gpu="{gid}"}}'] == 1
+    finally:
+        proc.terminate()
+        proc.wait(timeout=10)

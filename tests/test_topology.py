@@ -103,3 +103,30 @@ def test_unique_id_fallback_to_bdf(native_bin, tmp_path):
     props.write_text("\n".join(lines) + "\n")
     out = json.loads(run_smi(native_bin, root, "--json"))
     assert out["gpus"][0]["id"] == "amdgpu-0000:0c:00.0"
+
+
+def test_smi_health_columns(native_bin, tmp_path):
+    """mi355x-smi surfaces the same RAS/error counters the device
+    plugin's health model acts on (operator view == kubelet view)."""
+    import json
+    import subprocess
+    from sysfs_builder import make_gpu_sick
+    root = build_tree(tmp_path / "sys", n_gpus=2)
+    make_gpu_sick(root, 1, umc_ue=3, umc_ce=17, resets=1)
+    env = {"K3SAMD_SYSFS_ROOT": str(root)}
+    out = subprocess.run([str(native_bin / "mi355x-smi")], env=env,
+                         capture_output=True, text=True, timeout=60)
+    assert "ECC ue/ce" in out.stdout
+    assert "3/17" in out.stdout       # the sick GPU's counters
+    assert "n/a" in out.stdout        # GPU 0 has no RAS sysfs -> n/a
+    j = json.loads(subprocess.run(
+        [str(native_bin / "mi355x-smi"), "--json"], env=env,
+        capture_output=True, text=True, timeout=60).stdout)
+    sick = j["gpus"][1]
+    assert sick["ras_supported"] == "true"
+    assert sick["ras_uncorrectable"] == 3
+    assert sick["ras_correctable"] == 17
+    assert sick["reset_count"] == 1
+    healthy = j["gpus"][0]
+    assert healthy["ras_supported"] == "false"
+    assert healthy["ras_uncorrectable"] == -1
