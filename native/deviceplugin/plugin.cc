@@ -39,6 +39,12 @@ bool PluginConfig::from_yaml(const std::string& text, PluginConfig& out,
         out.health.max_uncorrectable = n->as_int(0);
       if (const YNode* n = h->get("maxCorrectableErrors"))
         out.health.max_correctable = n->as_int(10000);
+      if (const YNode* n = h->get("maxDeferredErrors"))
+        out.health.max_deferred = n->as_int(0);
+      if (const YNode* n = h->get("maxFatalEvents"))
+        out.health.max_fatal_events = n->as_int(0);
+      if (const YNode* n = h->get("maxBadPages"))
+        out.health.max_bad_pages = n->as_int(-1);
       if (const YNode* n = h->get("maxPcieReplays"))
         out.health.max_pcie_replays = n->as_int(-1);
       if (const YNode* n = h->get("maxResets"))

@@ -100,6 +100,9 @@ static int print_report(bool json) {
       w.key("ras_supported").value(hc.ras_present ? "true" : "false");
       w.key("ras_uncorrectable").value((int64_t)hc.ras_ue);
       w.key("ras_correctable").value((int64_t)hc.ras_ce);
+      w.key("ras_deferred").value((int64_t)hc.ras_de);
+      w.key("fatal_ras_events").value((int64_t)hc.fatal_events);
+      w.key("retired_vram_pages").value((int64_t)hc.bad_pages);
       w.key("pcie_replay_count").value((int64_t)hc.pcie_replay);
       w.key("reset_count").value((int64_t)hc.reset_count);
       w.end_obj();
@@ -110,8 +113,8 @@ static int print_report(bool json) {
     return 0;
   }
 
-  std::printf("+---------------------------------------------------------------------------+\n");
-  std::printf("| mi355x-smi            driver: %-12s                 k3samd stack   |\n",
+  std::printf("+-------------------------------------------------------------------------------+\n");
+  std::printf("| mi355x-smi            driver: %-25.25s      k3samd stack   |\n",
               topo.driver_version.empty() ? "unknown" : topo.driver_version.c_str());
   std::printf("+----+----------------------+--------+-----+-------------------+------+-----+------+-----------+\n");
   std::printf("| ## | Name                 | Arch   | CUs | VRAM used / total | xGMI | Tmp | Pwr  | ECC ue/ce |\n");

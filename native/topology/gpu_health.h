@@ -6,12 +6,19 @@
 // VANISHED; this module adds the sick-but-present signals so kubelet
 // stops scheduling pods onto a failing MI355X:
 //
-//   * RAS error counts:  <card>/device/ras/<block>_err_count files, each
-//     "ue: N\nce: N" (uncorrectable / correctable), summed over blocks
-//     (umc = HBM ECC, gfx, sdma, mmhub, ...).
-//   * PCIe replays:      <card>/device/pcie_replay_count (link health).
-//   * GPU resets:        <card>/device/reset_count or amdgpu reset_count
-//     (a recovered-from-hang GPU should drain before new pods land).
+//   * RAS error counts — both amdgpu generations (formats verified on a
+//     real MI355X, gpurun_out/r2_aca_probe.txt):
+//       - ACA banks (MI300+): <card>/device/ras/aca_<block> files, each
+//         "ue: N\nce: N\nde: N" (uncorrectable / correctable / deferred)
+//       - legacy:             <card>/device/ras/<block>_err_count files,
+//         "ue: N\nce: N"
+//   * Fatal RAS events:  ras/event_state ("Fatal Error: count:N, ...") —
+//     any fatal event means the GPU took (or needs) a reset.
+//   * Retired VRAM pages: ras/gpu_vram_bad_pages (one row per retired
+//     page; empty when clean).
+//   * PCIe replays:      <card>/device/pcie_replay_count (absent on the
+//     MI355X validation box — stays -1 there).
+//   * GPU resets:        <card>/device/reset_count (ditto).
 //
 // All reads are best-effort: a missing file yields -1 ("not exposed"),
 // which never trips a threshold — CPU fixtures and driver builds without
@@ -26,6 +33,9 @@ namespace k3samd {
 struct GpuHealthCounters {
   long ras_ue = -1;        // sum of uncorrectable errors over RAS blocks
   long ras_ce = -1;        // sum of correctable errors
+  long ras_de = -1;        // sum of deferred (poison-pending) errors, ACA
+  long fatal_events = -1;  // ras/event_state "Fatal Error" count
+  long bad_pages = -1;     // retired VRAM pages (rows in gpu_vram_bad_pages)
   long pcie_replay = -1;   // link-level replay count
   long reset_count = -1;   // completed GPU resets
   bool ras_present = false;
@@ -38,6 +48,10 @@ struct GpuHealthCounters {
 struct HealthPolicy {
   long max_uncorrectable = 0;
   long max_correctable = 10000;
+  long max_deferred = 0;       // deferred = poison pending consumption
+  long max_fatal_events = 0;   // any fatal RAS event ⇒ drain
+  long max_bad_pages = -1;     // retired pages are handled by the driver;
+                               // threshold only if the operator opts in
   long max_pcie_replays = -1;  // disabled by default (noisy on some hosts)
   long max_resets = 0;         // any completed reset ⇒ drain
 };

@@ -113,7 +113,11 @@ GpuRuntimeStats read_runtime_stats(const std::string& sysfs_root,
     while (struct dirent* e = ::readdir(d)) {
       if (e->d_name[0] == '.') continue;
       std::string base = hwdir + "/" + e->d_name;
-      if (st.temp_mc < 0) st.temp_mc = read_long(base + "/temp1_input", -1);
+      // sensor numbering varies: consumer cards expose temp1 (edge), the
+      // MI355X OAM exposes temp2 (junction) first — take the first present
+      for (int t = 1; t <= 3 && st.temp_mc < 0; ++t)
+        st.temp_mc =
+            read_long(base + "/temp" + std::to_string(t) + "_input", -1);
       if (st.power_uw < 0) {
         st.power_uw = read_long(base + "/power1_average", -1);
         if (st.power_uw < 0)

@@ -122,16 +122,37 @@ def build_tree(
 
 
 def make_gpu_sick(root, card_index, *, umc_ue=0, umc_ce=0, gfx_ue=0,
-                  gfx_ce=0, pcie_replays=None, resets=None):
+                  gfx_ce=0, pcie_replays=None, resets=None, aca=False,
+                  umc_de=0, fatal_events=None, bad_pages=None):
     """Write amdgpu RAS/error-state files for a present-but-sick GPU
-    (native/topology/gpu_health.cc reads these): per-block
-    `ras/<block>_err_count` ("ue: N\\nce: N"), `pcie_replay_count`,
-    `reset_count`. Call again with zeros to 'heal' the GPU."""
+    (native/topology/gpu_health.cc reads these). Two generations, both
+    verified against real sysfs:
+      legacy (aca=False): `ras/<block>_err_count` with "ue: N\\nce: N"
+      ACA    (aca=True):  `ras/aca_<block>` with "ue: N\\nce: N\\nde: N"
+                          (the MI355X format, gpurun_out/r2_aca_probe.txt)
+    plus `ras/event_state`, `ras/gpu_vram_bad_pages`,
+    `pcie_replay_count`, `reset_count`. Re-call with zeros to 'heal'."""
     dev = Path(root) / f"class/drm/card{card_index}/device"
     ras = dev / "ras"
     ras.mkdir(parents=True, exist_ok=True)
-    (ras / "umc_err_count").write_text(f"ue: {umc_ue}\nce: {umc_ce}\n")
-    (ras / "gfx_err_count").write_text(f"ue: {gfx_ue}\nce: {gfx_ce}\n")
+    if aca:
+        (ras / "aca_umc").write_text(
+            f"ue: {umc_ue}\nce: {umc_ce}\nde: {umc_de}\n")
+        (ras / "aca_gfx").write_text(
+            f"ue: {gfx_ue}\nce: {gfx_ce}\nde: 0\n")
+    else:
+        (ras / "umc_err_count").write_text(f"ue: {umc_ue}\nce: {umc_ce}\n")
+        (ras / "gfx_err_count").write_text(f"ue: {gfx_ue}\nce: {gfx_ce}\n")
+    if fatal_events is not None:
+        (ras / "event_state").write_text(
+            "current seqno: 7\n"
+            f"Fatal Error: count:{fatal_events}, last_seqno:0\n"
+            "Poison Creation: count:0, last_seqno:0\n"
+            "Poison Consumption: count:0, last_seqno:0\n")
+    if bad_pages is not None:
+        rows = "".join(f"0x{0x1000 * (i + 1):012x} : 0x1000 : R\n"
+                       for i in range(bad_pages))
+        (ras / "gpu_vram_bad_pages").write_text(rows)
     if pcie_replays is not None:
         (dev / "pcie_replay_count").write_text(f"{pcie_replays}\n")
     if resets is not None:

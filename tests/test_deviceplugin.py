@@ -332,6 +332,35 @@ def test_sick_gpu_reset_event_marked_unhealthy(harness):
     assert [d["health"] for d in second] == ["Unhealthy", "Unhealthy"]
 
 
+def test_sick_gpu_aca_format_marked_unhealthy(harness):
+    """The MI355X's actual RAS sysfs generation (ACA banks: ras/aca_umc
+    with ue:/ce:/de: lines — format captured from real silicon in
+    gpurun_out/r2_aca_probe.txt). A deferred (poison-pending) error must
+    flip the device Unhealthy under the default policy."""
+    from sysfs_builder import make_gpu_sick
+    h = harness(n_gpus=2, replicas=1, register=False, health_poll_ms=100)
+    stream = h.stream("ListAndWatch", timeout=30)
+    assert all(d["health"] == "Healthy"
+               for d in pb.decode_list_and_watch(next(stream)))
+    make_gpu_sick(h.root, card_index=0, aca=True, umc_de=1)
+    second = pb.decode_list_and_watch(next(stream))
+    health = {d["id"]: d["health"] for d in second}
+    assert sorted(health.values()) == ["Healthy", "Unhealthy"]
+    sick = [i for i, hl in health.items() if hl == "Unhealthy"][0]
+    assert sick.endswith("0000")
+
+
+def test_fatal_ras_event_marked_unhealthy(harness):
+    """ras/event_state 'Fatal Error: count:N' (the MI355X reset/poison
+    channel) drains the GPU."""
+    from sysfs_builder import make_gpu_sick
+    h = harness(n_gpus=1, replicas=1, register=False, health_poll_ms=100)
+    stream = h.stream("ListAndWatch", timeout=30)
+    assert pb.decode_list_and_watch(next(stream))[0]["health"] == "Healthy"
+    make_gpu_sick(h.root, card_index=0, aca=True, fatal_events=1)
+    assert pb.decode_list_and_watch(next(stream))[0]["health"] == "Unhealthy"
+
+
 def test_health_thresholds_configurable(harness):
     """Correctable-error budget comes from the additive `health:` config
     block; under the threshold stays Healthy, over it flips."""

@@ -83,3 +83,8 @@ def test_mi355x_smi_real_sysfs():
     assert g["arch"] == "gfx950"
     assert g["vram_bytes"] > 200 * (1 << 30)
     assert g["render_minor"] >= 128
+    # health/RAS fields present (values depend on the box's driver build;
+    # -1 = not exposed is acceptable, absence is not)
+    for key in ("ras_supported", "ras_uncorrectable", "ras_correctable",
+                "pcie_replay_count", "reset_count"):
+        assert key in g, key
