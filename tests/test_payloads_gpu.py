@@ -86,5 +86,12 @@ def test_mi355x_smi_real_sysfs():
     # health/RAS fields present (values depend on the box's driver build;
     # -1 = not exposed is acceptable, absence is not)
     for key in ("ras_supported", "ras_uncorrectable", "ras_correctable",
+                "ras_deferred", "fatal_ras_events", "retired_vram_pages",
                 "pcie_replay_count", "reset_count"):
         assert key in g, key
+    if g["ras_supported"] == "true":
+        # the MI355X pool exposes ACA RAS banks: counters must be real
+        # numbers (not the -1 'absent' sentinel) once RAS is present
+        assert g["ras_uncorrectable"] >= 0
+        assert g["ras_correctable"] >= 0
+        assert g["ras_deferred"] >= 0
