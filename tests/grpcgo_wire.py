@@ -387,28 +387,48 @@ class GrpcGoConn:
         """Next (kind, payload) event on want_stream; kind in
         {'headers', 'data', 'trailers', 'rst'}. Control frames handled
         transparently; DATA triggers a window-credit return (and
-        optionally a BDP ping), like grpc-go's transport."""
+        optionally a BDP ping), like grpc-go's transport. Events for
+        OTHER streams are buffered (grpc-go multiplexes concurrent RPCs
+        on one connection), so interleaved Allocate-during-ListAndWatch
+        flows demux correctly."""
+        pending = getattr(self, "_pending", None)
+        if pending is None:
+            pending = self._pending = {}
+        if pending.get(want_stream):
+            return pending[want_stream].pop(0)
         while True:
             ftype, flags, stream, payload = self.read_frame()
             if self.handle_control(ftype, flags, stream, payload):
                 continue
             if stream != want_stream:
+                ev = self._classify(ftype, flags, stream, payload,
+                                    bdp_ping_on_data=False)
+                if ev is not None:
+                    pending.setdefault(stream, []).append(ev)
                 continue
-            if ftype == DATA:
-                if payload:
-                    inc = struct.pack(">I", len(payload))
-                    self.send(frame(WINDOW_UPDATE, 0, stream, inc))
-                    self.send(frame(WINDOW_UPDATE, 0, 0, inc))
-                if bdp_ping_on_data:
-                    self.send(frame(PING, 0, 0, BDP_PING))
-                return ("data", payload)
-            if ftype == HEADERS:
-                hs = self.dec.decode(payload)
-                kind = "trailers" if any(n == "grpc-status" for n, _ in hs) \
-                    else "headers"
-                return (kind, hs)
-            if ftype == RST_STREAM:
-                return ("rst", payload)
+            ev = self._classify(ftype, flags, stream, payload,
+                                bdp_ping_on_data=bdp_ping_on_data)
+            if ev is not None:
+                return ev
+
+    def _classify(self, ftype, flags, stream, payload, *,
+                  bdp_ping_on_data):
+        if ftype == DATA:
+            if payload:
+                inc = struct.pack(">I", len(payload))
+                self.send(frame(WINDOW_UPDATE, 0, stream, inc))
+                self.send(frame(WINDOW_UPDATE, 0, 0, inc))
+            if bdp_ping_on_data:
+                self.send(frame(PING, 0, 0, BDP_PING))
+            return ("data", payload)
+        if ftype == HEADERS:
+            hs = self.dec.decode(payload)
+            kind = "trailers" if any(n == "grpc-status" for n, _ in hs) \
+                else "headers"
+            return (kind, hs)
+        if ftype == RST_STREAM:
+            return ("rst", payload)
+        return None
 
     def unary(self, path, body=b"", timeout_header=None, **ev_kw):
         sid = self.start_stream(path, body, timeout_header=timeout_header)

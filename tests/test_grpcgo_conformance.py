@@ -391,3 +391,121 @@ def test_plugin_client_registers_with_grpcgo_server(tmp_path):
         fake.stop()
         proc.terminate()
         proc.wait(timeout=10)
+
+
+def test_interleaved_allocate_during_listandwatch(plugin):
+    """kubelet's real concurrency pattern: ListAndWatch stays open while
+    Allocate RPCs multiplex on the SAME connection. Events must demux by
+    stream id — the Allocate response arrives while LW frames may be
+    in flight, and the LW stream keeps working afterwards."""
+    c = GrpcGoConn(plugin)
+    try:
+        lw = c.start_stream("/v1beta1.DevicePlugin/ListAndWatch")
+        kind, _ = c.read_stream_event(lw)
+        assert kind == "headers"
+        kind, first = c.read_stream_event(lw)
+        assert kind == "data"
+        devs = pb.decode_list_and_watch(first[5:])
+        # now, with LW open, run THREE Allocates on the same connection
+        for d in devs[:2] + [devs[0]]:
+            resp = pb.decode_allocate_response(
+                c.unary("/v1beta1.DevicePlugin/Allocate",
+                        pb.encode_allocate_request([[d["id"]]])))
+            assert d["id"] in resp[0]["envs"]["K3SAMD_VISIBLE_DEVICES"]
+        # LW must still be alive: cancel + reopen works on this conn
+        c.cancel(lw)
+        lw2 = c.start_stream("/v1beta1.DevicePlugin/ListAndWatch")
+        kind, _ = c.read_stream_event(lw2)
+        assert kind == "headers"
+        kind, data = c.read_stream_event(lw2)
+        assert len(pb.decode_list_and_watch(data[5:])) == len(devs)
+    finally:
+        c.close()
+
+
+def test_bytewise_tcp_fragmentation(plugin):
+    """TCP gives no framing guarantees: deliver an entire RPC one byte
+    per write. The server's frame reassembly must be position-agnostic."""
+    import socket as socket_mod
+    raw = socket_mod.socket(socket_mod.AF_UNIX, socket_mod.SOCK_STREAM)
+    raw.connect(plugin)
+    raw.settimeout(15)
+    enc = GoHpackEncoder()
+    block = enc.encode([(":method", "POST"), (":scheme", "http"),
+                        (":path", "/v1beta1.DevicePlugin/GetDevicePluginOptions"),
+                        (":authority", "x"),
+                        ("content-type", "application/grpc"),
+                        ("te", "trailers")])
+    wire = (PREFACE + frame(SETTINGS, 0, 0) +
+            frame(HEADERS, END_HEADERS, 1, block) +
+            frame(DATA, END_STREAM, 1, grpc_frame(b"")))
+    for i in range(len(wire)):
+        raw.sendall(wire[i:i + 1])
+    # read until trailers with grpc-status 0
+    buf = b""
+    status = None
+    while status is None:
+        chunk = raw.recv(65536)
+        assert chunk, "server closed on fragmented input"
+        buf += chunk
+        while len(buf) >= 9:
+            ln = int.from_bytes(buf[:3], "big")
+            if len(buf) < 9 + ln:
+                break
+            ftype, flags = buf[3], buf[4]
+            payload = buf[9:9 + ln]
+            buf = buf[9 + ln:]
+            if ftype == SETTINGS and not flags & ACK:
+                raw.sendall(frame(SETTINGS, ACK, 0))
+            elif ftype == HEADERS and b"grpc-status" in payload:
+                idx = payload.find(b"grpc-status")
+                vlen = payload[idx + 11]
+                status = int(payload[idx + 12: idx + 12 + vlen])
+    assert status == 0
+    raw.close()
+
+
+def test_conformance_dance_under_tsan(tmp_path):
+    """The grpc-go connection dance against the ThreadSanitizer build of
+    the plugin — the hand-rolled HTTP/2 threading under its strictest
+    client, with the race detector on."""
+    import shutil
+    tsan = PLUGIN.parent / "k3samd-device-plugin-tsan"
+    if not tsan.exists():
+        subprocess.run(["make", "-C", str(REPO / "native"), "tsan"],
+                       check=True, capture_output=True, timeout=600)
+    root = build_tree(tmp_path / "sys", n_gpus=2)
+    cfg = tmp_path / "config.yaml"
+    cfg.write_text(DEFAULT_CFG.format(replicas=2))
+    sock = str(tmp_path / "amd.sock")
+    proc = subprocess.Popen(
+        [str(tsan), "--config", str(cfg), "--plugin-sock", sock,
+         "--no-register", "--health-poll-ms", "50"],
+        env={"K3SAMD_SYSFS_ROOT": str(root),
+             "TSAN_OPTIONS": "exitcode=66 halt_on_error=0"},
+        stdout=subprocess.PIPE, stderr=subprocess.PIPE)
+    try:
+        deadline = time.time() + 15
+        while not Path(sock).exists():
+            assert time.time() < deadline and proc.poll() is None
+            time.sleep(0.05)
+        c = GrpcGoConn(sock)
+        lw = c.start_stream("/v1beta1.DevicePlugin/ListAndWatch")
+        assert c.read_stream_event(lw)[0] == "headers"
+        assert c.read_stream_event(lw, bdp_ping_on_data=True)[0] == "data"
+        assert c.wait_ping_ack(BDP_PING)
+        for _ in range(3):
+            c.unary("/v1beta1.DevicePlugin/GetDevicePluginOptions")
+        c.cancel(lw)
+        c.goaway_and_close()
+        time.sleep(0.3)
+    finally:
+        proc.terminate()
+        try:
+            proc.wait(timeout=15)
+        except subprocess.TimeoutExpired:
+            proc.kill()
+            proc.wait()
+    err = proc.stderr.read().decode(errors="replace")
+    assert "WARNING: ThreadSanitizer" not in err, err[-3000:]
+    assert proc.returncode != 66, "TSan reported races"
