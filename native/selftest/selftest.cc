@@ -287,6 +287,32 @@ int fuzz_main(const char* mode) {
     (void)dec.decode(input, out);
     return 0;
   }
+  if (!std::strcmp(mode, "hpack-dump")) {
+    // correctness oracle for encoder interop fuzzing: input is repeated
+    // [u32be len][header block]; all blocks decode through ONE decoder
+    // (dynamic-table state persists, as on a real connection); output is
+    // one "name\tvalue" line per header, "---" between blocks, "!err" on
+    // a decode failure.
+    k3samd::HpackDecoder dec;
+    size_t pos = 0;
+    while (pos + 4 <= input.size()) {
+      uint32_t len = ((uint32_t)(uint8_t)input[pos] << 24) |
+                     ((uint32_t)(uint8_t)input[pos + 1] << 16) |
+                     ((uint32_t)(uint8_t)input[pos + 2] << 8) |
+                     (uint8_t)input[pos + 3];
+      pos += 4;
+      if (pos + len > input.size()) break;
+      std::vector<k3samd::Header> out;
+      if (!dec.decode(input.substr(pos, len), out)) {
+        std::printf("!err\n");
+        return 1;
+      }
+      for (auto& [n, v] : out) std::printf("%s\t%s\n", n.c_str(), v.c_str());
+      std::printf("---\n");
+      pos += len;
+    }
+    return 0;
+  }
   if (!std::strcmp(mode, "huffman")) {
     std::string out;
     (void)k3samd::hpack_huffman_decode(input, out);
