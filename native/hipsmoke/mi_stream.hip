@@ -18,6 +18,7 @@
 #include <vector>
 
 #include "../../k3samd/ops/hip/stream_kernels.h"
+#include "../topology/gpu_health.h"
 #include "../topology/kfd_topology.h"
 
 #define HIP_CHECK(x)                                                         \
@@ -216,6 +217,10 @@ int main(int argc, char** argv) {
     }
     auto now = [] { return std::chrono::steady_clock::now(); };
     auto t_end = now() + std::chrono::seconds(burn_s);
+    // RAS snapshot before the stress: ECC errors that APPEAR during the
+    // burn are the node-acceptance failure signal
+    k3samd::GpuHealthCounters ras0 =
+        k3samd::read_gpu_health(k3samd::default_sysfs_root(), card);
     uint64_t triads = 0, mfmas = 0;
     long max_temp = -1, max_pw = -1;
     std::printf("burn: %d s of concurrent HBM streaming + bf16 MFMA\n",
@@ -263,12 +268,24 @@ int main(int argc, char** argv) {
     double avg_gbs = triads * 3.0 * buf_bytes / total_s / 1e9;
     double avg_tf =
         mfmas * mf_blocks * 4.0 * 4.0 * 16384.0 * 512.0 / total_s / 1e12;
+    k3samd::GpuHealthCounters ras1 =
+        k3samd::read_gpu_health(k3samd::default_sysfs_root(), card);
+    long d_ue = (ras0.ras_ue >= 0 && ras1.ras_ue >= 0)
+                    ? ras1.ras_ue - ras0.ras_ue : -1;
+    long d_ce = (ras0.ras_ce >= 0 && ras1.ras_ce >= 0)
+                    ? ras1.ras_ce - ras0.ras_ce : -1;
+    bool ras_clean = d_ue <= 0 && d_ce <= 0;
+    if (ras0.ras_present)
+      std::printf("burn RAS delta: ue %+ld ce %+ld (%s)\n", d_ue, d_ce,
+                  ras_clean ? "clean" : "ERRORS DURING BURN");
     std::printf(
         "{\"payload\": \"mi-burn\", \"seconds\": %.0f, \"avg_triad_gbps\": "
         "%.1f, \"avg_mfma_tflops\": %.1f, \"max_temp_c\": %ld, "
-        "\"max_power_w\": %.0f}\n",
+        "\"max_power_w\": %.0f, \"ras_delta_ue\": %ld, "
+        "\"ras_delta_ce\": %ld, \"ras_clean\": %s}\n",
         total_s, avg_gbs, avg_tf, max_temp < 0 ? -1 : max_temp / 1000,
-        max_pw < 0 ? -1.0 : max_pw / 1e6);
+        max_pw < 0 ? -1.0 : max_pw / 1e6, d_ue, d_ce,
+        ras_clean ? "true" : "false");
     HIP_CHECK(hipFree(mf_out));
     HIP_CHECK(hipStreamDestroy(s_mem));
     HIP_CHECK(hipStreamDestroy(s_mfma));
