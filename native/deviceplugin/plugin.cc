@@ -232,14 +232,44 @@ GrpcStatus DevicePlugin::handle_preferred(const std::string& req,
     std::vector<std::string> chosen(pr.must_include);
     std::set<std::string> used(chosen.begin(), chosen.end());
 
-    // Prefer packing replicas of the same physical GPU together (keeps
-    // other physical GPUs free for exclusive jobs), then same-NUMA spread.
+    // Two regimes:
+    //  * time-sliced (replicas>1): pack replicas of the same physical GPU
+    //    together — keeps other physical GPUs free for exclusive jobs.
+    //  * exclusive multi-GPU: prefer GPUs on the SAME host NUMA node as
+    //    the must-include set (or the node with the most free GPUs) —
+    //    xGMI is uniform all-to-all on the MI355X node, so the only
+    //    locality that differs is the host CPU/PCIe side feeding H2D.
     std::vector<std::string> avail;
     for (const auto& id : pr.available)
       if (!used.count(id)) avail.push_back(id);
+    int target_numa = -1;
+    if (cfg_.replicas <= 1) {
+      std::map<int, int> numa_free;
+      for (const auto& id : pr.must_include) {
+        int g = gpu_for_id(id);
+        if (g >= 0) target_numa = topo_.gpus[g].numa_node;
+      }
+      if (target_numa < 0) {
+        for (const auto& id : avail) {
+          int g = gpu_for_id(id);
+          if (g >= 0) numa_free[topo_.gpus[g].numa_node]++;
+        }
+        int best = -1;
+        for (auto& [node, cnt] : numa_free)
+          if (cnt > best) {
+            best = cnt;
+            target_numa = node;
+          }
+      }
+    }
     std::stable_sort(avail.begin(), avail.end(),
                      [&](const std::string& a, const std::string& b) {
                        int ga = gpu_for_id(a), gb = gpu_for_id(b);
+                       if (target_numa >= 0 && ga >= 0 && gb >= 0) {
+                         bool na = topo_.gpus[ga].numa_node == target_numa;
+                         bool nb = topo_.gpus[gb].numa_node == target_numa;
+                         if (na != nb) return na;
+                       }
                        if (ga != gb) return ga < gb;
                        return a < b;
                      });

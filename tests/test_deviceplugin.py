@@ -284,6 +284,39 @@ def test_preferred_allocation_packs_replicas(harness):
     assert len(bases) == 1
 
 
+def test_preferred_allocation_numa_affine_exclusive(harness):
+    """Exclusive multi-GPU requests prefer GPUs on the must-include set's
+    host NUMA node (fixture: even GPU indices on node 0, odd on node 1;
+    xGMI is uniform, host-side locality is what differs)."""
+    h = harness(n_gpus=8, replicas=1, register=False)
+    devs = pb.decode_list_and_watch(next(h.stream("ListAndWatch")))
+    avail = [d["id"] for d in devs]
+    must = [avail[1]]  # GPU 1 -> NUMA 1
+    resp = pb.decode_preferred_response(
+        h.call("GetPreferredAllocation",
+               pb.encode_preferred_request(avail, must, 4)))
+    chosen = resp[0]
+    assert len(chosen) == 4 and must[0] in chosen
+    # all four picks are odd-indexed GPUs (NUMA node 1)
+    idx = [int(c[-1], 16) for c in chosen]
+    assert all(i % 2 == 1 for i in idx), chosen
+
+
+def test_preferred_allocation_numa_majority_without_must(harness):
+    """With no must-include, pick the NUMA node that can satisfy the
+    request with the most locality: 3 free GPUs on node 0 vs 1 on node 1
+    -> a 3-GPU request lands entirely on node 0."""
+    h = harness(n_gpus=8, replicas=1, register=False)
+    devs = pb.decode_list_and_watch(next(h.stream("ListAndWatch")))
+    ids = [d["id"] for d in devs]
+    avail = [ids[0], ids[2], ids[4], ids[1]]  # three numa-0, one numa-1
+    resp = pb.decode_preferred_response(
+        h.call("GetPreferredAllocation",
+               pb.encode_preferred_request(avail, [], 3)))
+    idx = [int(c[-1], 16) for c in resp[0]]
+    assert sorted(idx) == [0, 2, 4], resp[0]
+
+
 def test_health_transition_pushes_update(harness):
     h = harness(n_gpus=2, replicas=1, register=False, health_poll_ms=100)
     stream = h.stream("ListAndWatch", timeout=30)
