@@ -40,16 +40,55 @@ static int print_report(bool json);
 int main(int argc, char** argv) {
   if (k3samd::handle_version_flag(argc, argv, "mi355x-smi")) return 0;
   bool json = false;
+  bool topo_matrix = false;
   int watch_s = 0;
   for (int i = 1; i < argc; ++i) {
     if (!std::strcmp(argv[i], "--json")) json = true;
+    else if (!std::strcmp(argv[i], "--topo")) topo_matrix = true;
     else if ((!std::strcmp(argv[i], "--watch") || !std::strcmp(argv[i], "-l"))
              && i + 1 < argc)
       watch_s = std::atoi(argv[++i]);
     else if (!std::strcmp(argv[i], "--help") || !std::strcmp(argv[i], "-h")) {
-      std::printf("usage: mi355x-smi [--json] [--watch SECONDS]\n");
+      std::printf(
+          "usage: mi355x-smi [--json] [--topo] [--watch SECONDS]\n");
       return 0;
     }
+  }
+  if (topo_matrix) {
+    // the `nvidia-smi topo -m` analog: pairwise connectivity from the
+    // KFD io_links. On a full MI355X node every pair is XGMI (7 direct
+    // point-to-point links per GPU); PHB = through the host bridge.
+    const std::string root = k3samd::default_sysfs_root();
+    k3samd::Topology topo = k3samd::enumerate_topology(root);
+    size_t n = topo.gpus.size();
+    if (n == 0) {
+      std::printf("no AMD GPUs found\n");
+      return 0;
+    }
+    std::printf("%8s", "");
+    for (size_t j = 0; j < n; ++j) std::printf(" %6s", ("GPU" + std::to_string(j)).c_str());
+    std::printf("   NUMA  BDF\n");
+    for (size_t i = 0; i < n; ++i) {
+      std::printf("%8s", ("GPU" + std::to_string(i)).c_str());
+      for (size_t j = 0; j < n; ++j) {
+        const char* cell = "PHB";
+        if (i == j) {
+          cell = "X";
+        } else {
+          for (int peer : topo.gpus[i].xgmi_peer_nodes)
+            if (peer == topo.gpus[j].kfd_node) {
+              cell = "XGMI";
+              break;
+            }
+        }
+        std::printf(" %6s", cell);
+      }
+      std::printf("  %4d   %s\n", topo.gpus[i].numa_node,
+                  topo.gpus[i].pci_bdf.c_str());
+    }
+    std::printf("\nLegend: XGMI = direct xGMI link, PHB = host bridge "
+                "path, X = self\n");
+    return 0;
   }
   if (watch_s > 0) {
     // refresh loop (the nvidia-smi -l analog); re-exec the table printer

@@ -200,6 +200,7 @@ Topology enumerate_topology(const std::string& sysfs_root) {
       uint64_t type = lp.count("type") ? lp["type"] : 0;
       if (type == kIoLinkXgmi) {
         g.xgmi_links++;
+        if (lp.count("node_to")) g.xgmi_peer_nodes.push_back((int)lp["node_to"]);
       } else if (type == kIoLinkPcie || type == 1 /*hypertransport*/) {
         if (lp.count("node_to")) g.numa_node = (int)lp["node_to"];
       }

@@ -37,6 +37,7 @@ struct GpuDevice {
   uint32_t simd_per_cu = 0;
   uint32_t gfx_target_version = 0; // e.g. 90500 => gfx950
   int xgmi_links = 0;              // count of type-11 io_links
+  std::vector<int> xgmi_peer_nodes;  // KFD node ids of xGMI peers
   int numa_node = 0;               // peer CPU node via io_links (best effort)
 
   uint32_t compute_units() const {

@@ -95,3 +95,15 @@ def test_mi355x_smi_real_sysfs():
         assert g["ras_uncorrectable"] >= 0
         assert g["ras_correctable"] >= 0
         assert g["ras_deferred"] >= 0
+
+
+@pytest.mark.gpu
+def test_mi355x_smi_topo_real():
+    """--topo connectivity matrix renders on the real box (single-GPU
+    lease: one row, no peer columns marked XGMI unless peers are
+    visible KFD nodes)."""
+    proc = subprocess.run([str(BIN / "mi355x-smi"), "--topo"],
+                          capture_output=True, text=True, timeout=60)
+    assert proc.returncode == 0, proc.stderr
+    assert "GPU0" in proc.stdout
+    assert "Legend" in proc.stdout

@@ -130,3 +130,28 @@ def test_smi_health_columns(native_bin, tmp_path):
     healthy = j["gpus"][0]
     assert healthy["ras_supported"] == "false"
     assert healthy["ras_uncorrectable"] == -1
+
+
+def test_smi_topo_matrix(native_bin, tmp_path):
+    """`mi355x-smi --topo` — the nvidia-smi `topo -m` analog: pairwise
+    xGMI connectivity from KFD io_links. The 4-GPU fixture is fully
+    connected; a 1-GPU tree has no peers."""
+    root = build_tree(tmp_path / "sys", n_gpus=4)
+    out = run_smi(native_bin, root, "--topo")
+    lines = [ln for ln in out.splitlines() if ln.strip().startswith("GPU")]
+    assert len(lines) == 4 + 0 or len(lines) >= 4
+    rows = [ln.split() for ln in out.splitlines()
+            if ln.strip().startswith("GPU") and "NUMA" not in ln]
+    assert len(rows) == 4
+    for i, row in enumerate(rows):
+        cells = row[1:5]
+        assert cells[i] == "X"
+        for j, c in enumerate(cells):
+            if j != i:
+                assert c == "XGMI", (i, j, row)
+    assert "0000:0c:00.0" in out  # BDF column
+    # single-GPU tree: no xGMI peers -> matrix is just the self cell
+    root1 = build_tree(tmp_path / "sys1", n_gpus=1)
+    out1 = run_smi(native_bin, root1, "--topo")
+    row = [ln for ln in out1.splitlines() if ln.strip().startswith("GPU0")][0]
+    assert "XGMI" not in row
