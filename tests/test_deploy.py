@@ -88,3 +88,17 @@ def test_chart_templates_exist():
     names = {p.name for p in (CHART / "templates").glob("*.yaml")}
     assert {"daemonset.yaml", "labeller-daemonset.yaml", "configmap.yaml",
             "runtimeclass.yaml"} <= names
+
+
+def test_mi_burn_job_contract():
+    """Node-acceptance Job: GPU contract knobs + no retries (a burn that
+    failed must be inspected, not silently re-run)."""
+    doc = load_docs(MANIFESTS / "mi-burn.yaml")[0]
+    assert doc["kind"] == "Job"
+    assert doc["spec"]["backoffLimit"] == 0
+    pod = doc["spec"]["template"]["spec"]
+    assert pod["runtimeClassName"] == "amd"
+    assert pod["restartPolicy"] == "Never"
+    c = pod["containers"][0]
+    assert c["resources"]["limits"]["amd.com/gpu"] == "1"
+    assert "--burn" in c["command"]
