@@ -38,15 +38,17 @@ void stream_triad(torch::Tensor a, torch::Tensor b, torch::Tensor c, double s,
   auto* ap = reinterpret_cast<f4*>(a.data_ptr<float>());
   auto* bp = reinterpret_cast<const f4*>(b.data_ptr<float>());
   auto* cp = reinterpret_cast<const f4*>(c.data_ptr<float>());
-  if (nontemporal) {
-    hipLaunchKernelGGL(k3samd_kern::stream_triad_kernel<true>,
-                       dim3(stream_grid(n4)), dim3(kThreadsPerBlock), 0,
-                       stream, ap, bp, cp, (float)s, n4);
-  } else {
-    hipLaunchKernelGGL(k3samd_kern::stream_triad_kernel<false>,
-                       dim3(stream_grid(n4)), dim3(kThreadsPerBlock), 0,
-                       stream, ap, bp, cp, (float)s, n4);
-  }
+  k3samd_kern::launch_chunked(n4, [&](int64_t off, int64_t cnt) {
+    if (nontemporal) {
+      hipLaunchKernelGGL(k3samd_kern::stream_triad_kernel<true>,
+                         dim3(stream_grid(cnt)), dim3(kThreadsPerBlock), 0,
+                         stream, ap + off, bp + off, cp + off, (float)s, cnt);
+    } else {
+      hipLaunchKernelGGL(k3samd_kern::stream_triad_kernel<false>,
+                         dim3(stream_grid(cnt)), dim3(kThreadsPerBlock), 0,
+                         stream, ap + off, bp + off, cp + off, (float)s, cnt);
+    }
+  });
   C10_HIP_KERNEL_LAUNCH_CHECK();
 }
 
@@ -58,15 +60,17 @@ void stream_copy(torch::Tensor a, torch::Tensor b, bool nontemporal) {
   auto stream = at::hip::getCurrentHIPStream();
   auto* ap = reinterpret_cast<f4*>(a.data_ptr<float>());
   auto* bp = reinterpret_cast<const f4*>(b.data_ptr<float>());
-  if (nontemporal) {
-    hipLaunchKernelGGL(k3samd_kern::stream_copy_kernel<true>,
-                       dim3(stream_grid(n4)), dim3(kThreadsPerBlock), 0,
-                       stream, ap, bp, n4);
-  } else {
-    hipLaunchKernelGGL(k3samd_kern::stream_copy_kernel<false>,
-                       dim3(stream_grid(n4)), dim3(kThreadsPerBlock), 0,
-                       stream, ap, bp, n4);
-  }
+  k3samd_kern::launch_chunked(n4, [&](int64_t off, int64_t cnt) {
+    if (nontemporal) {
+      hipLaunchKernelGGL(k3samd_kern::stream_copy_kernel<true>,
+                         dim3(stream_grid(cnt)), dim3(kThreadsPerBlock), 0,
+                         stream, ap + off, bp + off, cnt);
+    } else {
+      hipLaunchKernelGGL(k3samd_kern::stream_copy_kernel<false>,
+                         dim3(stream_grid(cnt)), dim3(kThreadsPerBlock), 0,
+                         stream, ap + off, bp + off, cnt);
+    }
+  });
   C10_HIP_KERNEL_LAUNCH_CHECK();
 }
 
@@ -78,15 +82,17 @@ void stream_scale(torch::Tensor a, torch::Tensor c, double s, bool nontemporal) 
   auto stream = at::hip::getCurrentHIPStream();
   auto* ap = reinterpret_cast<f4*>(a.data_ptr<float>());
   auto* cp = reinterpret_cast<const f4*>(c.data_ptr<float>());
-  if (nontemporal) {
-    hipLaunchKernelGGL(k3samd_kern::stream_scale_kernel<true>,
-                       dim3(stream_grid(n4)), dim3(kThreadsPerBlock), 0,
-                       stream, ap, cp, (float)s, n4);
-  } else {
-    hipLaunchKernelGGL(k3samd_kern::stream_scale_kernel<false>,
-                       dim3(stream_grid(n4)), dim3(kThreadsPerBlock), 0,
-                       stream, ap, cp, (float)s, n4);
-  }
+  k3samd_kern::launch_chunked(n4, [&](int64_t off, int64_t cnt) {
+    if (nontemporal) {
+      hipLaunchKernelGGL(k3samd_kern::stream_scale_kernel<true>,
+                         dim3(stream_grid(cnt)), dim3(kThreadsPerBlock), 0,
+                         stream, ap + off, cp + off, (float)s, cnt);
+    } else {
+      hipLaunchKernelGGL(k3samd_kern::stream_scale_kernel<false>,
+                         dim3(stream_grid(cnt)), dim3(kThreadsPerBlock), 0,
+                         stream, ap + off, cp + off, (float)s, cnt);
+    }
+  });
   C10_HIP_KERNEL_LAUNCH_CHECK();
 }
 
@@ -101,15 +107,17 @@ void stream_add(torch::Tensor a, torch::Tensor b, torch::Tensor c,
   auto* ap = reinterpret_cast<f4*>(a.data_ptr<float>());
   auto* bp = reinterpret_cast<const f4*>(b.data_ptr<float>());
   auto* cp = reinterpret_cast<const f4*>(c.data_ptr<float>());
-  if (nontemporal) {
-    hipLaunchKernelGGL(k3samd_kern::stream_add_kernel<true>,
-                       dim3(stream_grid(n4)), dim3(kThreadsPerBlock), 0,
-                       stream, ap, bp, cp, n4);
-  } else {
-    hipLaunchKernelGGL(k3samd_kern::stream_add_kernel<false>,
-                       dim3(stream_grid(n4)), dim3(kThreadsPerBlock), 0,
-                       stream, ap, bp, cp, n4);
-  }
+  k3samd_kern::launch_chunked(n4, [&](int64_t off, int64_t cnt) {
+    if (nontemporal) {
+      hipLaunchKernelGGL(k3samd_kern::stream_add_kernel<true>,
+                         dim3(stream_grid(cnt)), dim3(kThreadsPerBlock), 0,
+                         stream, ap + off, bp + off, cp + off, cnt);
+    } else {
+      hipLaunchKernelGGL(k3samd_kern::stream_add_kernel<false>,
+                         dim3(stream_grid(cnt)), dim3(kThreadsPerBlock), 0,
+                         stream, ap + off, bp + off, cp + off, cnt);
+    }
+  });
   C10_HIP_KERNEL_LAUNCH_CHECK();
 }
 

@@ -253,4 +253,25 @@ inline int64_t stream_grid(int64_t n4) {
   return (n4 + kThreadsPerBlock - 1) / kThreadsPerBlock;
 }
 
+// The AQL dispatch packet's grid_size_x is a 32-bit WORK-ITEM count, so a
+// single flat launch caps at 2^32-1 threads — a 64 GiB fp32 buffer (2^32
+// float4s) needs exactly 2^32 and fails with hipErrorInvalidConfiguration
+// (hit in practice at buffer-MiB >= 65536; gpurun_out/r2_buffer_sweep.err).
+// Launch in <= kMaxFlatChunk pieces instead of switching to grid-stride:
+// the flat kernel beat every grid-stride tuning (profiles/
+// r01_triad_tuning_sweep.txt) and a second launch per 64 GiB is free.
+constexpr int64_t kMaxFlatChunk =
+    (int64_t(4294967295u) / kThreadsPerBlock) * kThreadsPerBlock;
+
+template <typename LaunchOne>
+inline void launch_chunked(int64_t n4, LaunchOne&& launch_one) {
+  int64_t off = 0;
+  do {
+    int64_t count = n4 - off;
+    if (count > kMaxFlatChunk) count = kMaxFlatChunk;
+    launch_one(off, count);
+    off += count;
+  } while (off < n4);
+}
+
 }  // namespace k3samd_kern

@@ -103,8 +103,12 @@ int main(int argc, char** argv) {
     auto launch_all = [&] {
       for (int d = 0; d < ndev; ++d) {
         (void)hipSetDevice(d);
-        hipLaunchKernelGGL(k3samd_kern::stream_triad_kernel<true>, g, blk, 0,
-                           streams[d], A[d], B[d], C[d], 2.5f, n4_);
+        k3samd_kern::launch_chunked(n4_, [&](int64_t off, int64_t cnt) {
+          dim3 gc((uint32_t)k3samd_kern::stream_grid(cnt));
+          hipLaunchKernelGGL(k3samd_kern::stream_triad_kernel<true>, gc, blk,
+                             0, streams[d], A[d] + off, B[d] + off,
+                             C[d] + off, 2.5f, cnt);
+        });
       }
     };
     auto sync_all = [&] {
@@ -307,21 +311,26 @@ int main(int argc, char** argv) {
 
   auto launch = [&](int op, bool nt) {
     using namespace k3samd_kern;
-    if (nt) {
-      switch (op) {
-        case 0: hipLaunchKernelGGL(stream_copy_kernel<true>, grid, block, 0, 0, a, b, n4); break;
-        case 1: hipLaunchKernelGGL(stream_scale_kernel<true>, grid, block, 0, 0, a, c, s, n4); break;
-        case 2: hipLaunchKernelGGL(stream_add_kernel<true>, grid, block, 0, 0, a, b, c, n4); break;
-        case 3: hipLaunchKernelGGL(stream_triad_kernel<true>, grid, block, 0, 0, a, b, c, s, n4); break;
+    // chunked: a flat launch caps at 2^32-1 work-items (AQL grid_size_x),
+    // so buffers >= 64 GiB need more than one dispatch
+    launch_chunked(n4, [&](int64_t off, int64_t cnt) {
+      dim3 g((uint32_t)stream_grid(cnt));
+      if (nt) {
+        switch (op) {
+          case 0: hipLaunchKernelGGL(stream_copy_kernel<true>, g, block, 0, 0, a + off, b + off, cnt); break;
+          case 1: hipLaunchKernelGGL(stream_scale_kernel<true>, g, block, 0, 0, a + off, c + off, s, cnt); break;
+          case 2: hipLaunchKernelGGL(stream_add_kernel<true>, g, block, 0, 0, a + off, b + off, c + off, cnt); break;
+          case 3: hipLaunchKernelGGL(stream_triad_kernel<true>, g, block, 0, 0, a + off, b + off, c + off, s, cnt); break;
+        }
+      } else {
+        switch (op) {
+          case 0: hipLaunchKernelGGL(stream_copy_kernel<false>, g, block, 0, 0, a + off, b + off, cnt); break;
+          case 1: hipLaunchKernelGGL(stream_scale_kernel<false>, g, block, 0, 0, a + off, c + off, s, cnt); break;
+          case 2: hipLaunchKernelGGL(stream_add_kernel<false>, g, block, 0, 0, a + off, b + off, c + off, cnt); break;
+          case 3: hipLaunchKernelGGL(stream_triad_kernel<false>, g, block, 0, 0, a + off, b + off, c + off, s, cnt); break;
+        }
       }
-    } else {
-      switch (op) {
-        case 0: hipLaunchKernelGGL(stream_copy_kernel<false>, grid, block, 0, 0, a, b, n4); break;
-        case 1: hipLaunchKernelGGL(stream_scale_kernel<false>, grid, block, 0, 0, a, c, s, n4); break;
-        case 2: hipLaunchKernelGGL(stream_add_kernel<false>, grid, block, 0, 0, a, b, c, n4); break;
-        case 3: hipLaunchKernelGGL(stream_triad_kernel<false>, grid, block, 0, 0, a, b, c, s, n4); break;
-      }
-    }
+    });
   };
 
   for (int op = 0; op < 4; ++op) {
