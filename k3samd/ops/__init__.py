@@ -66,3 +66,14 @@ MFMA_LAYOUT = 0
 def mfma_gemm16(A, B, layout=MFMA_LAYOUT):
     """Single-tile D[16,16] = A[16,32] @ B[32,16] via one bf16 MFMA."""
     return _require_native().mfma_gemm16(A, B, int(layout))
+
+
+def mx_gemm16(A, B, fmt=0):
+    """Single-tile D[16,16] = dequant(A[16,128]) @ dequant(B[128,16]) via
+    one block-scaled v_mfma_scale_f32_16x16x128_f8f6f4 with unit scales.
+
+    fmt 0 = fp8 e4m3 (A uint8 [16,128]; B uint8 [16,128] packed
+    column-major: row c = column c of B). fmt 4 = fp4 e2m1 (packed two
+    elements per byte, low nibble = even k: A [16,64], B [16,64]).
+    """
+    return _require_native().mx_gemm16(A, B, int(fmt))

@@ -177,6 +177,41 @@ torch::Tensor mfma_gemm16(torch::Tensor A, torch::Tensor B, int64_t layout) {
   return D;
 }
 
+// Block-scaled MX single-tile GEMM: D[16,16] = A[16,128] x B[128,16]
+// with fp8-e4m3 (fmt=0) or fp4-e2m1 (fmt=4) operands, unit scales.
+// A: uint8 [16,128] (fp8) / [16,64] (fp4, 2 elems per byte, low nibble =
+// even k). B: uint8 packed COLUMN-major with k contiguous: [16,128]
+// (fp8: row c holds column c of B) / [16,64] (fp4).
+torch::Tensor mx_gemm16(torch::Tensor A, torch::Tensor B, int64_t fmt) {
+  K3_CHECK(A.is_cuda() && B.is_cuda(), "A,B must be on GPU");
+  K3_CHECK(A.scalar_type() == torch::kUInt8 &&
+               B.scalar_type() == torch::kUInt8,
+           "A,B must be uint8 (packed fp8/fp4)");
+  K3_CHECK(A.is_contiguous() && B.is_contiguous(), "A,B must be contiguous");
+  K3_CHECK(fmt == 0 || fmt == 4, "fmt must be 0 (fp8 e4m3) or 4 (fp4)");
+  int64_t kb = fmt == 0 ? 128 : 64;  // packed bytes along k
+  K3_CHECK(A.size(0) == 16 && A.size(1) == kb && B.size(0) == 16 &&
+               B.size(1) == kb,
+           "A must be [16,KB], B [16,KB] (B column-major-packed)");
+  auto D = torch::empty({16, 16}, A.options().dtype(torch::kFloat32));
+  auto stream = at::hip::getCurrentHIPStream();
+  if (fmt == 0) {
+    hipLaunchKernelGGL(k3samd_kern::mx_gemm16_kernel<0>, dim3(1), dim3(64),
+                       0, stream,
+                       reinterpret_cast<const uint8_t*>(A.data_ptr()),
+                       reinterpret_cast<const uint8_t*>(B.data_ptr()),
+                       D.data_ptr<float>());
+  } else {
+    hipLaunchKernelGGL(k3samd_kern::mx_gemm16_kernel<4>, dim3(1), dim3(64),
+                       0, stream,
+                       reinterpret_cast<const uint8_t*>(A.data_ptr()),
+                       reinterpret_cast<const uint8_t*>(B.data_ptr()),
+                       D.data_ptr<float>());
+  }
+  C10_HIP_KERNEL_LAUNCH_CHECK();
+  return D;
+}
+
 // Host-side query: the extension only ships gfx950 code objects, so report
 // whether the active device is gfx950.
 bool has_mfma() {
@@ -201,5 +236,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("shape") = 16);
   m.def("mfma_gemm16", &mfma_gemm16, py::arg("A"), py::arg("B"),
         py::arg("layout") = 0);
+  m.def("mx_gemm16", &mx_gemm16, py::arg("A"), py::arg("B"),
+        py::arg("fmt") = 0);
   m.def("has_mfma", &has_mfma);
 }
