@@ -46,6 +46,13 @@ def build_extension(verbose: bool = True) -> Path:
     if not _needs_rebuild():
         return so
 
+    # torch's generated build.ninja tracks only the .hip source, NOT the
+    # included headers — a header-only change leaves ninja convinced the
+    # object is fresh and silently re-ships a stale kernel (this bit us:
+    # an empty mx_gemm16_kernel survived two rebuilds). Touch the source
+    # so ninja always recompiles when we decided a rebuild is needed.
+    HIP_SRC.touch()
+
     os.environ.setdefault("PYTORCH_ROCM_ARCH", GFX_ARCH)
     from torch.utils.cpp_extension import load
 

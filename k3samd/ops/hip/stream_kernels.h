@@ -79,60 +79,6 @@ __global__ void stream_add_kernel(f4* __restrict__ a, const f4* __restrict__ b,
   }
 }
 
-// Single-tile D[16x16] = dequant(A[16x128]) * dequant(B[128x16]) through
-// one block-scaled v_mfma_scale_f32_16x16x128_f8f6f4 — numerics
-// validation for the MX path (the throughput kernels above never check
-// values). Scales fixed at E8M0 1.0 so dequant is the identity on the
-// element encodings.
-//
-// Presumed A/B fragment layout (generalizing the silicon-validated bf16
-// 16x16x32 map — 4 lanes per row, contiguous k-chunk per lane):
-//   lane l holds A[l&15][(l>>4)*32 + j]  for j = 0..31 (fp8: byte j of
-//   the 32-byte operand; fp4: nibble j of the 16-byte operand, low
-//   nibble first), B mirrored with row/col swapped.
-// C/D use the shape-determined map col=lane&15, row=(lane>>4)*4+reg
-// (dtype-independent per the CDNA4 ISA).
-//
-// A is packed row-major (fp8: 1 B/elem, stride 128; fp4: 2 elems/B, low
-// nibble = even k, stride 64 B). B is packed COLUMN-major with k
-// contiguous (fp8: B_bytes[col*128 + k]; fp4: nibbles along k,
-// B_bytes[col*64 + k/2]) so both gathers are contiguous per lane.
-template <int FMT>  // 0 = fp8 e4m3, 4 = fp4 e2m1
-static __global__ void mx_gemm16_kernel(const uint8_t* __restrict__ A,
-                                        const uint8_t* __restrict__ B,
-                                        float* __restrict__ D) {
-#if K3_HAS_MFMA
-  const int l = threadIdx.x;  // one wave
-  i32x8 a = {0, 0, 0, 0, 0, 0, 0, 0}, b = {0, 0, 0, 0, 0, 0, 0, 0};
-  uint8_t* ab = reinterpret_cast<uint8_t*>(&a);
-  uint8_t* bb = reinterpret_cast<uint8_t*>(&b);
-  if constexpr (FMT == 0) {
-#pragma unroll
-    for (int j = 0; j < 32; ++j) {
-      int k = (l >> 4) * 32 + j;
-      ab[j] = A[(l & 15) * 128 + k];
-      bb[j] = B[(l & 15) * 128 + k];  // B packed col-major: col=l&15
-    }
-  } else {  // fp4: 32 nibbles = 16 bytes per lane
-#pragma unroll
-    for (int j = 0; j < 16; ++j) {
-      int kb = (l >> 4) * 16 + j;  // byte index along the packed k axis
-      ab[j] = A[(l & 15) * 64 + kb];
-      bb[j] = B[(l & 15) * 64 + kb];
-    }
-  }
-  const int scale = 0x7f7f7f7f;  // E8M0 1.0
-  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-  acc = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
-      a, b, acc, FMT, FMT, 0, scale, 0, scale);
-#pragma unroll
-  for (int r = 0; r < 4; ++r)
-    D[((l >> 4) * 4 + r) * 16 + (l & 15)] = acc[r];
-#else
-  (void)A; (void)B; (void)D;
-#endif
-}
-
 // Grid-stride triad with U float4-elements per thread per pass — used by
 // the tuning sweep (mi-stream --tune) to probe block-size/unroll space.
 template <bool NT, int U>
@@ -300,6 +246,66 @@ static __global__ void mfma_gemm16_kernel(const uint16_t* __restrict__ A,
   }
 #else
   (void)A; (void)B; (void)D; (void)layout;
+#endif
+}
+
+// Single-tile D[16x16] = dequant(A[16x128]) * dequant(B[128x16]) through
+// one block-scaled v_mfma_scale_f32_16x16x128_f8f6f4 — numerics
+// validation for the MX path (the throughput kernels above never check
+// values). Scales fixed at E8M0 1.0 so dequant is the identity on the
+// element encodings.
+//
+// Presumed A/B fragment layout (generalizing the silicon-validated bf16
+// 16x16x32 map — 4 lanes per row, contiguous k-chunk per lane):
+//   lane l holds A[l&15][(l>>4)*32 + j]  for j = 0..31 (fp8: byte j of
+//   the 32-byte operand; fp4: nibble j of the 16-byte operand, low
+//   nibble first), B mirrored with row/col swapped.
+// C/D use the shape-determined map col=lane&15, row=(lane>>4)*4+reg
+// (dtype-independent per the CDNA4 ISA).
+//
+// A is packed row-major (fp8: 1 B/elem, stride 128; fp4: 2 elems/B, low
+// nibble = even k, stride 64 B). B is packed COLUMN-major with k
+// contiguous (fp8: B_bytes[col*128 + k]; fp4: nibbles along k,
+// B_bytes[col*64 + k/2]) so both gathers are contiguous per lane.
+template <int FMT>  // 0 = fp8 e4m3, 4 = fp4 e2m1
+static __global__ void mx_gemm16_kernel(const uint8_t* __restrict__ A,
+                                        const uint8_t* __restrict__ B,
+                                        float* __restrict__ D) {
+#if K3_HAS_MFMA
+  const int l = threadIdx.x;  // one wave
+  i32x8 a = {0, 0, 0, 0, 0, 0, 0, 0}, b = {0, 0, 0, 0, 0, 0, 0, 0};
+  uint8_t* ab = reinterpret_cast<uint8_t*>(&a);
+  uint8_t* bb = reinterpret_cast<uint8_t*>(&b);
+  if constexpr (FMT == 0) {
+#pragma unroll
+    for (int j = 0; j < 32; ++j) {
+      int k = (l >> 4) * 32 + j;
+      ab[j] = A[(l & 15) * 128 + k];
+      bb[j] = B[(l & 15) * 128 + k];  // B packed col-major: col=l&15
+    }
+  } else {  // fp4: 32 nibbles = 16 bytes per lane
+#pragma unroll
+    for (int j = 0; j < 16; ++j) {
+      int kb = (l >> 4) * 16 + j;  // byte index along the packed k axis
+      ab[j] = A[(l & 15) * 64 + kb];
+      bb[j] = B[(l & 15) * 64 + kb];
+    }
+  }
+  const int scale = 0x7f7f7f7f;  // E8M0 1.0
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  acc = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+      a, b, acc, FMT, FMT, 0, scale, 0, scale);
+  // Defensive: keep the scale VGPR alive past the MFMA. Without this,
+  // LLVM allocates the destination v[0:3] OVER the scale source
+  // (verified in the generated asm); MFMA sources are read across the
+  // instruction's internal passes and the ISA's overlap rules for this
+  // operand class are not documented — don't rely on it.
+  __asm__ volatile("" ::"v"(scale));
+#pragma unroll
+  for (int r = 0; r < 4; ++r)
+    D[((l >> 4) * 4 + r) * 16 + (l & 15)] = acc[r];
+#else
+  (void)A; (void)B; (void)D;
 #endif
 }
 
