@@ -132,7 +132,9 @@ double mfma_throughput(torch::Tensor out, int64_t iters, int64_t shape) {
   int64_t blocks = out.numel();
   K3_CHECK(blocks > 0 && blocks <= (1 << 22), "bad block count");
   auto stream = at::hip::getCurrentHIPStream();
-  double flops_per_wave_iter = 4.0 * 16384.0;  // shapes 16/32 issue equal FLOPs
+  // shape 16: 4 chains x 16384 FLOP per iter; shape 32: 4 chains x 2
+  // unroll per i+=2 step = 4 MFMAs/iter x 32768 FLOP
+  double flops_per_wave_iter = shape == 32 ? 4.0 * 32768.0 : 4.0 * 16384.0;
   if (shape == 16) {
     hipLaunchKernelGGL(k3samd_kern::mfma_throughput_kernel, dim3(blocks),
                        dim3(kThreadsPerBlock), 0, stream,

@@ -166,16 +166,22 @@ static __global__ void mfma_throughput32_kernel(float* __restrict__ out,
     a[j] = (short)(0x3f80 + ((threadIdx.x + j) & 7));
     b[j] = (short)(0x3f00 + ((threadIdx.x * 3 + j) & 7));
   }
-  f32x16 acc0 = {}, acc1 = {};
-  // 2 chains x 2-deep unroll: hides loop scalar overhead behind the
-  // 64-cycle MFMA issue slots
+  // 4 chains x 2-deep unroll at high grid occupancy: the r02 sweep
+  // (tools/mfma_tune.hip, gpurun_out/r2_mfma_tune.txt) measured 2477 TF
+  // with ch=4 at 4096 blocks vs 2203 TF for the old 2-chain config —
+  // 99 % of the 2.5 PF dense spec peak.
+  f32x16 acc0 = {}, acc1 = {}, acc2 = {}, acc3 = {};
   for (int i = 0; i < iters; i += 2) {
     acc0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc0, 0, 0, 0);
     acc1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc1, 0, 0, 0);
+    acc2 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc2, 0, 0, 0);
+    acc3 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc3, 0, 0, 0);
     acc0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc0, 0, 0, 0);
     acc1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc1, 0, 0, 0);
+    acc2 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc2, 0, 0, 0);
+    acc3 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc3, 0, 0, 0);
   }
-  float r = acc0[0] + acc1[1];
+  float r = acc0[0] + acc1[1] + acc2[2] + acc3[3];
   if (threadIdx.x == 0) out[blockIdx.x] = r;
 #else
   if (threadIdx.x == 0) out[blockIdx.x] = -1.f;

@@ -362,25 +362,31 @@ int main(int argc, char** argv) {
 
   double mfma_tf = 0, mfma32_tf = 0;
   if (do_mfma) {
-    const int blocks = 2048, mf_iters = 4096;
+    // 4096 blocks + 4 chains measured best (tools/mfma_tune.hip sweep:
+    // 2477 TF = 99 % of the 2.5 PF dense peak)
+    const int blocks = 4096, mf_iters = 4096;
     float* out;
     HIP_CHECK(hipMalloc(&out, blocks * sizeof(float)));
     auto time_mfma = [&](auto kern, double flops_per_wave_iter,
                          int accs) -> double {
       hipLaunchKernelGGL(kern, dim3(blocks), block, 0, 0, out, 256);  // warm
       (void)hipDeviceSynchronize();
-      (void)hipEventRecord(ev0);
-      hipLaunchKernelGGL(kern, dim3(blocks), block, 0, 0, out, mf_iters);
-      (void)hipEventRecord(ev1);
-      (void)hipEventSynchronize(ev1);
-      float ms;
-      (void)hipEventElapsedTime(&ms, ev0, ev1);
+      float best = 1e30f;
+      for (int rep = 0; rep < 3; ++rep) {  // best-of-3: ride out clock ramp
+        (void)hipEventRecord(ev0);
+        hipLaunchKernelGGL(kern, dim3(blocks), block, 0, 0, out, mf_iters);
+        (void)hipEventRecord(ev1);
+        (void)hipEventSynchronize(ev1);
+        float ms;
+        (void)hipEventElapsedTime(&ms, ev0, ev1);
+        if (ms < best) best = ms;
+      }
       double flops =
           (double)blocks * 4 * accs * flops_per_wave_iter * mf_iters;
-      return flops / (ms * 1e9);
+      return flops / (best * 1e9);
     };
     mfma_tf = time_mfma(k3samd_kern::mfma_throughput_kernel, 16384.0, 4);
-    mfma32_tf = time_mfma(k3samd_kern::mfma_throughput32_kernel, 32768.0, 2);
+    mfma32_tf = time_mfma(k3samd_kern::mfma_throughput32_kernel, 32768.0, 4);
     double mx8_tf =
         time_mfma(k3samd_kern::mfma_throughput_mx_kernel<0>, 65536.0, 4);
     double mx6_tf =
