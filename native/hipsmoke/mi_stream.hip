@@ -12,6 +12,7 @@
 // Build: hipcc --offload-arch=gfx950 -O3 mi_stream.hip -o mi-stream
 
 #include <chrono>
+#include <cmath>
 #include <cstdio>
 #include <cstring>
 #include <string>
@@ -41,6 +42,123 @@ struct Timing {
 
 double gbps(double bytes, float ms) { return bytes / (ms * 1e6); }
 
+// Numerics self-check: the pod log doubles as a correctness oracle (the
+// reference's nvidia-smi output carried version oracles the same way,
+// /root/reference/README.md:139-147). Three checks against host-computed
+// references: fp32 triad (exact), one bf16 MFMA tile, one block-scaled
+// MX-fp8 tile (identity-A => must reproduce B exactly).
+bool self_check() {
+  using namespace k3samd_kern;
+  bool ok = true;
+  // --- triad, 4096 elements, exactly-representable values
+  {
+    const int64_t n = 4096, n4 = n / 4;
+    float *a, *b, *c;
+    (void)hipMalloc(&a, n * 4);
+    (void)hipMalloc(&b, n * 4);
+    (void)hipMalloc(&c, n * 4);
+    std::vector<float> hb(n), hc(n), ha(n);
+    for (int64_t i = 0; i < n; ++i) {
+      hb[i] = (float)((i % 31) - 15);
+      hc[i] = (float)((i % 17) - 8) * 0.5f;
+    }
+    (void)hipMemcpy(b, hb.data(), n * 4, hipMemcpyHostToDevice);
+    (void)hipMemcpy(c, hc.data(), n * 4, hipMemcpyHostToDevice);
+    hipLaunchKernelGGL(stream_triad_kernel<true>, dim3(stream_grid(n4)),
+                       dim3(kThreadsPerBlock), 0, 0, (f4*)a, (f4*)b, (f4*)c,
+                       2.5f, n4);
+    (void)hipMemcpy(ha.data(), a, n * 4, hipMemcpyDeviceToHost);
+    for (int64_t i = 0; i < n && ok; ++i)
+      if (ha[i] != hb[i] + 2.5f * hc[i]) ok = false;
+    (void)hipFree(a); (void)hipFree(b); (void)hipFree(c);
+    if (!ok) { std::printf("numerics: triad FAIL\n"); return false; }
+  }
+#if 1
+  // --- bf16 MFMA tile vs host float reference
+  double mfma_err = 0;
+  {
+    uint16_t *A, *B; float* D;
+    (void)hipMalloc(&A, 16 * 32 * 2);
+    (void)hipMalloc(&B, 32 * 16 * 2);
+    (void)hipMalloc(&D, 16 * 16 * 4);
+    std::vector<uint16_t> hA(16 * 32), hB(32 * 16);
+    std::vector<float> fA(16 * 32), fB(32 * 16);
+    auto to_bf16 = [](float f) -> uint16_t {
+      uint32_t u;
+      std::memcpy(&u, &f, 4);
+      return (uint16_t)(u >> 16);  // values chosen exactly representable
+    };
+    for (int i = 0; i < 16 * 32; ++i) {
+      fA[i] = (float)((i % 13) - 6) * 0.25f;
+      hA[i] = to_bf16(fA[i]);
+    }
+    for (int i = 0; i < 32 * 16; ++i) {
+      fB[i] = (float)((i % 11) - 5) * 0.5f;   // asymmetric vs A
+      hB[i] = to_bf16(fB[i]);
+    }
+    (void)hipMemcpy(A, hA.data(), hA.size() * 2, hipMemcpyHostToDevice);
+    (void)hipMemcpy(B, hB.data(), hB.size() * 2, hipMemcpyHostToDevice);
+    hipLaunchKernelGGL(mfma_gemm16_kernel, dim3(1), dim3(64), 0, 0, A, B, D,
+                       0);
+    std::vector<float> hD(256);
+    (void)hipMemcpy(hD.data(), D, 256 * 4, hipMemcpyDeviceToHost);
+    for (int i = 0; i < 16; ++i)
+      for (int j = 0; j < 16; ++j) {
+        float ref = 0;
+        for (int k = 0; k < 32; ++k) ref += fA[i * 32 + k] * fB[k * 16 + j];
+        double e = std::abs(hD[i * 16 + j] - ref);
+        if (e > mfma_err) mfma_err = e;
+      }
+    (void)hipFree(A); (void)hipFree(B); (void)hipFree(D);
+    if (mfma_err > 1e-3) {
+      std::printf("numerics: bf16 MFMA FAIL (err %g)\n", mfma_err);
+      return false;
+    }
+  }
+  // --- MX-fp8 identity tile: D must equal B rows exactly
+  double mx_err = 0;
+  {
+    uint8_t *A, *B; float* D;
+    (void)hipMalloc(&A, 16 * 128);
+    (void)hipMalloc(&B, 16 * 128);
+    (void)hipMalloc(&D, 16 * 16 * 4);
+    std::vector<uint8_t> hA(16 * 128, 0), hB(16 * 128);
+    for (int i = 0; i < 16; ++i) hA[i * 128 + i] = 0x38;  // e4m3 1.0 on diag
+    // B values: exact small e4m3 codes, col-major pack, asymmetric
+    std::vector<float> fB(128 * 16);
+    const float lut[8] = {0.f, 0.5f, 1.f, 1.5f, 2.f, 3.f, 4.f, 6.f};
+    for (int k = 0; k < 128; ++k)
+      for (int j = 0; j < 16; ++j) {
+        int code = (k * 7 + j * 3) % 16;  // uses sign bit too
+        float v = lut[code & 7] * ((code & 8) ? -1.f : 1.f);
+        fB[k * 16 + j] = v;
+        // e4m3 encodings of the lut: 0,0x30,0x38,0x3c,0x40,0x44,0x48,0x4c
+        const uint8_t enc[8] = {0, 0x30, 0x38, 0x3c, 0x40, 0x44, 0x48, 0x4c};
+        hB[j * 128 + k] = enc[code & 7] | ((code & 8) ? 0x80 : 0);
+      }
+    (void)hipMemcpy(A, hA.data(), hA.size(), hipMemcpyHostToDevice);
+    (void)hipMemcpy(B, hB.data(), hB.size(), hipMemcpyHostToDevice);
+    hipLaunchKernelGGL(mx_gemm16_kernel<0>, dim3(1), dim3(64), 0, 0, A, B, D);
+    std::vector<float> hD(256);
+    (void)hipMemcpy(hD.data(), D, 256 * 4, hipMemcpyDeviceToHost);
+    for (int i = 0; i < 16; ++i)
+      for (int j = 0; j < 16; ++j) {
+        double e = std::abs(hD[i * 16 + j] - fB[i * 16 + j]);
+        if (e > mx_err) mx_err = e;
+      }
+    (void)hipFree(A); (void)hipFree(B); (void)hipFree(D);
+    if (mx_err != 0.0) {
+      std::printf("numerics: MX-fp8 FAIL (err %g)\n", mx_err);
+      return false;
+    }
+  }
+  std::printf(
+      "numerics: triad exact, bf16 MFMA err %.2g, MX-fp8 identity exact\n",
+      mfma_err);
+#endif
+  return true;
+}
+
 }  // namespace
 
 int main(int argc, char** argv) {
@@ -48,6 +166,7 @@ int main(int argc, char** argv) {
   int iters = 20;
   int device = 0;
   bool do_mfma = true;
+  bool do_check = true;
   bool tune = false;
   bool all_gpus = false;
   int burn_s = 0;
@@ -60,6 +179,8 @@ int main(int argc, char** argv) {
       device = std::atoi(argv[++i]);
     else if (!std::strcmp(argv[i], "--no-mfma"))
       do_mfma = false;
+    else if (!std::strcmp(argv[i], "--no-check"))
+      do_check = false;
     else if (!std::strcmp(argv[i], "--tune"))
       tune = true;
     else if (!std::strcmp(argv[i], "--all-gpus"))
@@ -145,6 +266,8 @@ int main(int argc, char** argv) {
   std::printf("mi-stream: %s (%s), %d visible GPU(s), %d CUs, %.0f GiB VRAM\n",
               gpu_name, prop.gcnArchName, ndev, prop.multiProcessorCount,
               (double)prop.totalGlobalMem / (1 << 30));
+
+  if (do_check && !self_check()) return 3;  // pod log is the oracle
 
   const int64_t n = mib * (1 << 20) / 4;  // fp32 elements
   const int64_t n4 = n / 4;

@@ -24,6 +24,9 @@ def test_mi_stream_runs():
                            "--iters", "5"],
                           capture_output=True, text=True, timeout=300)
     assert proc.returncode == 0, proc.stderr
+    # built-in numerics oracle must have run and passed (exit 3 otherwise)
+    assert "numerics: triad exact" in proc.stdout
+    assert "MX-fp8 identity exact" in proc.stdout
     last = proc.stdout.strip().splitlines()[-1]
     j = json.loads(last)
     assert j["payload"] == "mi-stream"
