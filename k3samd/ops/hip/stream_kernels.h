@@ -114,6 +114,41 @@ __global__ void stream_triad_gs_kernel(f4* __restrict__ a,
   for (; i < n4; i += stride) a[i] = triad_op(b[i], c[i], s);
 }
 
+// LDS-staged triad (SURVEY §2c's blueprint variant, kept for the measured
+// comparison): stage b and c tiles into LDS with the async 16-byte
+// `global_load_lds` DMA, then compute from LDS. For a pure streaming op
+// with ZERO reuse this adds an LDS round trip per element, so the direct
+// non-temporal kernel above is expected to win — mi-stream --lds-compare
+// measures the gap rather than asserting it. The LDS destination of
+// global_load_lds is wave-uniform base + lane*16 (per-lane scatter is
+// not a thing), so each wave stages its own contiguous 64-f4 slice.
+__global__ void stream_triad_lds_kernel(f4* __restrict__ a,
+                                        const f4* __restrict__ b,
+                                        const f4* __restrict__ c, float s,
+                                        int64_t n4) {
+#if defined(__gfx950__)
+  __shared__ f4 sb[kThreadsPerBlock], sc[kThreadsPerBlock];
+  const int tid = threadIdx.x;
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + tid;
+  // lanes past the end stage element 0 (the whole wave must issue the
+  // DMA together); their results are simply not stored
+  int64_t i_ld = i < n4 ? i : 0;
+  f4* wave_sb = sb + (tid & ~63);  // wave-uniform LDS base
+  f4* wave_sc = sc + (tid & ~63);
+  __builtin_amdgcn_global_load_lds((const __attribute__((address_space(1))) void*)&b[i_ld],
+                                   (__attribute__((address_space(3))) void*)wave_sb,
+                                   16, 0, 0);
+  __builtin_amdgcn_global_load_lds((const __attribute__((address_space(1))) void*)&c[i_ld],
+                                   (__attribute__((address_space(3))) void*)wave_sc,
+                                   16, 0, 0);
+  __syncthreads();  // drains the in-flight DMA (vmcnt(0) before barrier)
+  if (i >= n4) return;
+  __builtin_nontemporal_store(triad_op(sb[tid], sc[tid], s), &a[i]);
+#else
+  (void)a; (void)b; (void)c; (void)s; (void)n4;
+#endif
+}
+
 // ---------------------------------------------------------------------------
 // MFMA kernels (device code only selected when compiling for gfx950).
 // ---------------------------------------------------------------------------

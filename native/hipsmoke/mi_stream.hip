@@ -168,6 +168,7 @@ int main(int argc, char** argv) {
   bool do_mfma = true;
   bool do_check = true;
   bool tune = false;
+  bool lds_compare = false;
   bool all_gpus = false;
   int burn_s = 0;
   for (int i = 1; i < argc; ++i) {
@@ -183,13 +184,15 @@ int main(int argc, char** argv) {
       do_check = false;
     else if (!std::strcmp(argv[i], "--tune"))
       tune = true;
+    else if (!std::strcmp(argv[i], "--lds-compare"))
+      lds_compare = true;
     else if (!std::strcmp(argv[i], "--all-gpus"))
       all_gpus = true;
     else if (!std::strcmp(argv[i], "--burn") && i + 1 < argc)
       burn_s = std::atoi(argv[++i]);
     else {
       std::printf("mi-stream [--mib N] [--iters N] [--device D] [--no-mfma]"
-                  " [--tune] [--all-gpus] [--burn SECONDS]\n");
+                  " [--tune] [--lds-compare] [--all-gpus] [--burn SECONDS]\n");
       return !std::strcmp(argv[i], "--help") ? 0 : 2;
     }
   }
@@ -285,6 +288,41 @@ int main(int argc, char** argv) {
   dim3 grid((uint32_t)k3samd_kern::stream_grid(n4));
   dim3 block(k3samd_kern::kThreadsPerBlock);
   const float s = 2.5f;
+
+  if (lds_compare) {
+    // direct-nt vs LDS-staged triad (SURVEY §2c blueprint comparison)
+    const double bytes = 3 * buf_bytes;
+    auto time_k = [&](auto launch_fn) -> double {
+      launch_fn();
+      (void)hipDeviceSynchronize();
+      float best = 1e30f;
+      for (int it = 0; it < iters; ++it) {
+        (void)hipEventRecord(ev0);
+        launch_fn();
+        (void)hipEventRecord(ev1);
+        (void)hipEventSynchronize(ev1);
+        float ms;
+        (void)hipEventElapsedTime(&ms, ev0, ev1);
+        if (ms < best) best = ms;
+      }
+      return gbps(bytes, best);
+    };
+    double direct = time_k([&] {
+      hipLaunchKernelGGL(k3samd_kern::stream_triad_kernel<true>, grid, block,
+                         0, 0, a, b, c, s, n4);
+    });
+    double lds = time_k([&] {
+      hipLaunchKernelGGL(k3samd_kern::stream_triad_lds_kernel, grid, block,
+                         0, 0, a, b, c, s, n4);
+    });
+    std::printf("triad %lld MiB: direct-nt %.1f GB/s | LDS-staged %.1f GB/s "
+                "(%.1f%%)\n",
+                (long long)mib, direct, lds, 100.0 * lds / direct);
+    std::printf("{\"payload\": \"mi-stream-lds\", \"direct_gbps\": %.1f, "
+                "\"lds_staged_gbps\": %.1f}\n", direct, lds);
+    (void)hipFree(a); (void)hipFree(b); (void)hipFree(c);
+    return 0;
+  }
 
   if (tune) {
     // sweep block-size x unroll x grid-occupancy for the grid-stride
