@@ -44,6 +44,8 @@ int main(int argc, char** argv) {
   int health_poll_ms = 5000;
   bool oneshot = false;
   bool use_cdi = false;
+  int register_retries = 5;
+  int register_backoff_ms = 1000;
   std::string metrics_addr;
 
   for (int i = 1; i < argc; ++i) {
@@ -56,6 +58,8 @@ int main(int argc, char** argv) {
     else if (const char* v = arg("--kubelet-sock")) kubelet_sock = v;
     else if (const char* v = arg("--health-poll-ms")) health_poll_ms = std::atoi(v);
     else if (!std::strcmp(argv[i], "--no-register")) kubelet_sock.clear();
+    else if (const char* v = arg("--register-retries")) register_retries = std::atoi(v);
+    else if (const char* v = arg("--register-backoff-ms")) register_backoff_ms = std::atoi(v);
     else if (!std::strcmp(argv[i], "--use-cdi")) use_cdi = true;
     else if (const char* v = arg("--metrics-addr")) metrics_addr = v;
     else if (!std::strcmp(argv[i], "--oneshot")) oneshot = true;
@@ -82,6 +86,7 @@ int main(int argc, char** argv) {
 
   cfg.use_cdi = use_cdi;
   k3samd::DevicePlugin plugin(cfg, k3samd::default_sysfs_root());
+  plugin.set_register_policy(register_retries, register_backoff_ms);
   if (!config_path.empty()) plugin.watch_config(config_path);
 
   if (oneshot) {

@@ -516,7 +516,23 @@ bool DevicePlugin::serve(const std::string& plugin_sock,
   stopping_.store(false);
 
   if (!kubelet_sock.empty()) {
-    if (!register_with_kubelet()) {
+    // kubelet may come up after this DaemonSet pod (node boot ordering):
+    // retry with linear backoff before giving up to CrashLoop semantics.
+    bool registered = false;
+    for (int attempt = 0; attempt <= register_retries_; ++attempt) {
+      if (attempt > 0) {
+        std::fprintf(stderr,
+                     "deviceplugin: Register retry %d/%d in %d ms\n",
+                     attempt, register_retries_, register_backoff_ms_);
+        std::this_thread::sleep_for(
+            std::chrono::milliseconds(register_backoff_ms_));
+      }
+      if (register_with_kubelet()) {
+        registered = true;
+        break;
+      }
+    }
+    if (!registered) {
       server_.stop();
       return false;
     }

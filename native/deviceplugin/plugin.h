@@ -56,6 +56,12 @@ class DevicePlugin {
   // Re-enumerate the topology and rebuild the advertised device list.
   void refresh_topology();
 
+  // kubelet-registration retry policy (kubelet may start after us).
+  void set_register_policy(int retries, int backoff_ms) {
+    register_retries_ = retries;
+    register_backoff_ms_ = backoff_ms;
+  }
+
   // Watch `path` for config changes (mtime/size) from the health loop;
   // a valid new config rebuilds the device list (pushed over
   // ListAndWatch); an invalid one is logged and ignored.
@@ -111,6 +117,8 @@ class DevicePlugin {
   MetricsServer metrics_server_;
   std::thread health_thread_;
   std::thread reregister_thread_;
+  int register_retries_ = 5;       // extra attempts after the first
+  int register_backoff_ms_ = 1000;
   std::string plugin_sock_;
   std::string kubelet_sock_;
   std::atomic<bool> stopping_{false};

@@ -530,3 +530,45 @@ def test_bad_config_rejected(tmp_path):
                          capture_output=True, text=True, timeout=60)
     assert out.returncode == 2
     assert "version" in out.stderr
+
+
+def test_register_retries_until_kubelet_appears(tmp_path):
+    """Node boot ordering: the plugin DaemonSet may start before kubelet's
+    device-plugin registry socket exists. The plugin must retry with
+    backoff and register once the (fake) kubelet comes up."""
+    root = build_tree(tmp_path / "sys", n_gpus=1)
+    kubelet_sock = str(tmp_path / "kubelet.sock")
+    proc = subprocess.Popen(
+        [str(PLUGIN), "--plugin-sock", str(tmp_path / "amd.sock"),
+         "--kubelet-sock", kubelet_sock,
+         "--register-retries", "20", "--register-backoff-ms", "200",
+         "--health-poll-ms", "0"],
+        env={"K3SAMD_SYSFS_ROOT": str(root)},
+        stdout=subprocess.PIPE, stderr=subprocess.PIPE)
+    kubelet = None
+    try:
+        time.sleep(0.8)  # a few failed attempts happen first
+        assert proc.poll() is None, proc.stderr.read().decode()
+        kubelet = FakeKubelet(kubelet_sock)
+        assert kubelet.event.wait(10), "plugin never registered"
+        assert kubelet.requests[0]["resource_name"] == "amd.com/gpu"
+        time.sleep(0.2)
+        assert proc.poll() is None  # serving normally
+    finally:
+        if kubelet:
+            kubelet.stop()
+        proc.terminate()
+        proc.wait(timeout=10)
+
+
+def test_register_gives_up_after_retries(tmp_path):
+    """Exhausted retries still end in a loud exit (DaemonSet backoff)."""
+    root = build_tree(tmp_path / "sys", n_gpus=1)
+    proc = subprocess.run(
+        [str(PLUGIN), "--plugin-sock", str(tmp_path / "amd.sock"),
+         "--kubelet-sock", str(tmp_path / "missing.sock"),
+         "--register-retries", "2", "--register-backoff-ms", "50"],
+        env={"K3SAMD_SYSFS_ROOT": str(root)},
+        capture_output=True, text=True, timeout=60)
+    assert proc.returncode == 1
+    assert "Register retry 2/2" in proc.stderr
