@@ -126,6 +126,18 @@ def main(argv=None) -> int:
         b *= 4
     rows = busbw_sweep(sizes, iters=args.iters, device=device, op=args.op)
 
+    # correctness verdict: rank r contributes (r+1); every element must be
+    # world*(world+1)/2 on every rank
+    vt = torch.full((1 << 12,), float(rank + 1),
+                    device=device if device is not None else "cpu")
+    dist.all_reduce(vt)
+    expect = world * (world + 1) / 2.0
+    verified = bool((vt == expect).all().item())
+    vflag = torch.tensor([1 if verified else 0],
+                         device=device if device is not None else "cpu")
+    dist.all_reduce(vflag, op=dist.ReduceOp.MIN)
+    verified = bool(int(vflag[0]) == 1)
+
     transports = None
     if rccl_log is not None:
         from k3samd.utils.rccl_diag import gather_transport_counts
@@ -136,7 +148,8 @@ def main(argv=None) -> int:
             print(f"{r['bytes']:>12d} B  algbw {r['algbw_gbps']:8.2f} GB/s"
                   f"  busbw {r['busbw_gbps']:8.2f} GB/s")
         result = {"payload": "allreduce", "backend": backend,
-                  "op": args.op, "world_size": world, "rows": rows}
+                  "op": args.op, "world_size": world, "rows": rows,
+                  "verify_ok": verified}
         if backend == "nccl" and world > 1:
             # xGMI sanity verdict: the fully-connected
             # MI355X node must beat one xGMI link's ~153 GB/s at large
@@ -152,7 +165,7 @@ def main(argv=None) -> int:
                     and non_p2p == 0
         print(json.dumps(result))
     dist.destroy_process_group()
-    return 0
+    return 0 if verified else 4
 
 
 if __name__ == "__main__":

@@ -138,6 +138,41 @@ int main(int argc, char** argv) {
     }
   }
   std::printf("+-----------+------------+------------+\n");
+
+  // Correctness verdict: rank i contributes the constant (i+1), so every
+  // element of every rank's result must equal n(n+1)/2 — the first 8-GPU
+  // contact verifies itself instead of only timing itself.
+  bool verify_ok = true;
+  {
+    const size_t vcount = 1 << 20;
+    float expect = (float)(ngpus * (ngpus + 1)) / 2.0f;
+    for (int i = 0; i < ngpus; ++i) {
+      HIP_CHECK(hipSetDevice(i));
+      float fill = (float)(i + 1);
+      uint32_t bits;
+      std::memcpy(&bits, &fill, 4);
+      HIP_CHECK(hipMemsetD32((hipDeviceptr_t)sendb[i], (int)bits, vcount));
+    }
+    ncclGroupStart();
+    for (int i = 0; i < ngpus; ++i)
+      ncclAllReduce(sendb[i], recvb[i], vcount, ncclFloat, ncclSum, comms[i],
+                    streams[i]);
+    ncclGroupEnd();
+    for (int i = 0; i < ngpus; ++i) HIP_CHECK(hipStreamSynchronize(streams[i]));
+    std::vector<float> probe(3);
+    for (int i = 0; i < ngpus && verify_ok; ++i) {
+      HIP_CHECK(hipSetDevice(i));
+      float* checks[3] = {recvb[i], recvb[i] + vcount / 2,
+                          recvb[i] + vcount - 1};
+      for (int p = 0; p < 3; ++p) {
+        HIP_CHECK(hipMemcpy(&probe[p], checks[p], 4, hipMemcpyDeviceToHost));
+        if (probe[p] != expect) verify_ok = false;
+      }
+    }
+    std::printf("allreduce verify (sum of ranks, expect %.0f): %s\n", expect,
+                verify_ok ? "PASS" : "FAIL");
+  }
+
   // xGMI sanity verdict: on the fully-connected MI355X node
   // a healthy multi-GPU all-reduce must beat ONE xGMI link's ~153 GB/s —
   // proof that RCCL drives parallel p2p paths through the injected
@@ -151,8 +186,10 @@ int main(int argc, char** argv) {
   }
   std::printf(
       "{\"payload\": \"mi-allreduce\", \"n_gpus\": %d, \"max_busbw_gbps\": "
-      "%.1f, \"xgmi_p2p_ok\": %s}\n",
-      ngpus, last_busbw, xgmi_ok ? "true" : "false");
+      "%.1f, \"xgmi_p2p_ok\": %s, \"verify_ok\": %s}\n",
+      ngpus, last_busbw, xgmi_ok ? "true" : "false",
+      verify_ok ? "true" : "false");
+  if (!verify_ok) return 4;
 
   for (int i = 0; i < ngpus; ++i) {
     ncclCommDestroy(comms[i]);

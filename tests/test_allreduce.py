@@ -112,3 +112,21 @@ def test_torchrun_world4_cli(tmp_path):
     j = json.loads(last[0])
     assert j["world_size"] == 4
     assert j["rows"][0]["busbw_gbps"] > 0
+
+
+def test_verify_flag_in_output(tmp_path):
+    """The payload's JSON carries the self-verification verdict (rank r
+    contributes r+1; every element must equal world*(world+1)/2)."""
+    proc = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run",
+         "--nnodes=1", "--nproc-per-node=2",
+         "--master-addr", "127.0.0.1", "--master-port", free_port(),
+         "-m", "k3samd.parallel.allreduce",
+         "--backend", "gloo", "--min-mib", "1", "--max-mib", "1",
+         "--iters", "2"],
+        cwd=str(REPO), env=dict(os.environ), capture_output=True, text=True,
+        timeout=600)
+    assert proc.returncode == 0, proc.stderr[-2000:]
+    j = json.loads([ln for ln in proc.stdout.splitlines()
+                    if ln.startswith("{")][0])
+    assert j["verify_ok"] is True
