@@ -42,7 +42,7 @@ fi
 # GPU enumeration (our topology lib view)
 smi=$(command -v mi355x-smi || echo ./native/bin/mi355x-smi)
 if [ -x "$smi" ]; then
-  gpus=$("$smi" --json 2>/dev/null | sed -n 's/.*"gpu_count": \([0-9]*\).*/\1/p')
+  gpus=$("$smi" --json 2>/dev/null | sed -n 's/.*"gpu_count": *\([0-9]*\).*/\1/p')
   if [ "${gpus:-0}" -ge 1 ]; then
     pass "mi355x-smi enumerates $gpus GPU(s)"
   else

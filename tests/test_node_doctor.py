@@ -44,6 +44,11 @@ def run_doctor(tmp_path, *, kfd=True, renders=2, containerd=True,
     rt = bindir / "k3samd-oci-runtime"
     rt.write_text("#!/bin/sh\nexit 0\n")
     rt.chmod(rt.stat().st_mode | stat.S_IEXEC)
+    # fake mi355x-smi emitting the real (space-free) JSON shape — the
+    # real box caught a sed that assumed a space after the colon
+    smi = bindir / "mi355x-smi"
+    smi.write_text('#!/bin/sh\necho \'{"driver_version":"x","gpu_count":2,"gpus":[]}\'\n')
+    smi.chmod(smi.stat().st_mode | stat.S_IEXEC)
     env = dict(os.environ)
     env.update({
         "PATH": f"{bindir}:{env['PATH']}",
@@ -63,6 +68,7 @@ def test_ready_node_passes(tmp_path):
     assert proc.returncode == 0, proc.stdout
     assert "node ready." in proc.stdout
     assert "FAIL" not in proc.stdout
+    assert "enumerates 2 GPU(s)" in proc.stdout
 
 
 def test_missing_kfd_fails(tmp_path):
