@@ -86,3 +86,41 @@ def test_runtimeclass_renders():
     assert rc["kind"] == "RuntimeClass"
     assert rc["handler"] == "amd"
     assert rc["metadata"]["name"] == "amd"
+
+
+def test_configmap_health_block_flows_through():
+    """The additive health: thresholds reach the mounted config intact
+    (the plugin's PluginConfig parser consumes exactly this document)."""
+    (cm,) = render_template("configmap.yaml")
+    cfg = yaml.safe_load(cm["data"]["config.yaml"])
+    h = cfg["health"]
+    assert h["maxUncorrectableErrors"] == 0
+    assert h["maxCorrectableErrors"] == 10000
+    assert h["maxDeferredErrors"] == 0
+    assert h["maxFatalEvents"] == 0
+    assert h["maxBadPages"] == -1
+    assert h["maxResets"] == 0
+
+
+def test_values_matrix_renders_valid_objects():
+    """Sweep the user-facing knobs; every template must render to valid
+    objects (or nothing, for gated templates) in every combination."""
+    combos = [
+        {},
+        {"gfd": {"enabled": False}},
+        {"cdi": {"enabled": True}},
+        {"metrics": {"enabled": False}},
+        {"runtimeClassName": "amd-exp"},
+        {"gfd": {"enabled": False}, "cdi": {"enabled": True},
+         "metrics": {"enabled": False}},
+    ]
+    for overrides in combos:
+        for tpl in ("daemonset.yaml", "configmap.yaml",
+                    "labeller-daemonset.yaml", "runtimeclass.yaml",
+                    "servicemonitor.yaml"):
+            docs = render_template(tpl, overrides)
+            for d in docs:
+                assert d.get("kind"), (tpl, overrides)
+                assert d.get("apiVersion"), (tpl, overrides)
+                meta = d.get("metadata", {})
+                assert meta.get("name"), (tpl, overrides)
