@@ -366,6 +366,7 @@ constexpr int64_t kMaxFlatChunk =
 
 template <typename LaunchOne>
 inline void launch_chunked(int64_t n4, LaunchOne&& launch_one) {
+  if (n4 <= 0) return;  // a 0-block grid is an invalid launch
   int64_t off = 0;
   do {
     int64_t count = n4 - off;
