@@ -81,6 +81,14 @@ void MetricsServer::serve_loop() {
     timeval tv{2, 0};
     ::setsockopt(cfd, SOL_SOCKET, SO_RCVTIMEO, &tv, sizeof(tv));
     ::setsockopt(cfd, SOL_SOCKET, SO_SNDTIMEO, &tv, sizeof(tv));
+    // bound concurrent connection threads: a flood (the endpoint may be
+    // bound on the pod IP for ServiceMonitor scraping) must not grow
+    // threads without limit — excess connections are dropped, and the
+    // 2 s timeouts guarantee the pool drains
+    if (active_.load() >= 64) {
+      ::close(cfd);
+      continue;
+    }
     active_.fetch_add(1);
     std::thread([this, cfd] {
       char buf[1024];

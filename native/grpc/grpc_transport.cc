@@ -366,6 +366,13 @@ void GrpcServer::serve_loop() {
     conn->fd = cfd;
     {
       std::lock_guard<std::mutex> lk(conns_mu_);
+      // kubelet holds ONE connection for the plugin's lifetime; a cap
+      // far above that bounds thread growth if something floods the
+      // socket (each connection owns a reader thread)
+      if (conns_.size() >= 32) {
+        ::close(cfd);
+        continue;
+      }
       conns_.insert(conn);
     }
     spawn([this, conn] { connection_loop(conn); });

@@ -83,3 +83,39 @@ def test_many_listandwatch_streams(tmp_path):
         assert opts["get_preferred_allocation_available"]
     finally:
         h.close()
+
+
+def test_connection_flood_bounded(tmp_path):
+    """A flood of idle connections must not grow plugin threads without
+    bound: excess connections beyond the cap are dropped, and the
+    (single) legitimate kubelet connection keeps working."""
+    import socket as socket_mod
+    from test_deviceplugin import PluginHarness
+    import pb_v1beta1 as pb
+    h = PluginHarness(tmp_path, n_gpus=1, replicas=1, register=False)
+    # grpcio connects lazily: make the legitimate connection FIRST so it
+    # owns a slot before the flood arrives (kubelet's long-lived
+    # connection predates any attack in practice)
+    opts = pb.decode_options(h.call("GetDevicePluginOptions", timeout=10))
+    assert opts["get_preferred_allocation_available"] is True
+    floods = []
+    try:
+        for _ in range(60):
+            try:
+                s = socket_mod.socket(socket_mod.AF_UNIX,
+                                      socket_mod.SOCK_STREAM)
+                s.connect(h.plugin_sock)
+                floods.append(s)
+            except OSError:
+                break
+        time.sleep(0.3)
+        # the established client still gets served mid-flood
+        opts = pb.decode_options(h.call("GetDevicePluginOptions", timeout=10))
+        assert opts["get_preferred_allocation_available"] is True
+        # thread count is bounded: /proc/<pid>/task
+        tasks = len(list(Path(f"/proc/{h.proc.pid}/task").iterdir()))
+        assert tasks < 64, f"{tasks} threads after flood"
+    finally:
+        for s in floods:
+            s.close()
+        h.close()
