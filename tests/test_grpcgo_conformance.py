@@ -509,3 +509,115 @@ def test_conformance_dance_under_tsan(tmp_path):
     err = proc.stderr.read().decode(errors="replace")
     assert "WARNING: ThreadSanitizer" not in err, err[-3000:]
     assert proc.returncode != 66, "TSan reported races"
+
+
+class FlakyGrpcGoKubelet:
+    """First Register gets a grpc-go TRAILERS-ONLY error response (the
+    wire shape grpc-go uses for immediate errors: one HEADERS frame with
+    :status + grpc-status, END_STREAM, no DATA); the second connection
+    succeeds. Exercises the plugin client's trailers-only parsing AND
+    its retry loop."""
+
+    def __init__(self, sock_path):
+        self.path = sock_path
+        self.register_ok = threading.Event()
+        self.attempts = 0
+        self.srv = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+        Path(sock_path).unlink(missing_ok=True)
+        self.srv.bind(sock_path)
+        self.srv.listen(4)
+        self.thread = threading.Thread(target=self._serve, daemon=True)
+        self.thread.start()
+
+    def _one(self, conn, fail):
+        conn.settimeout(10)
+        enc = GoHpackEncoder()
+        buf = b""
+
+        def read_frame():
+            nonlocal buf
+            while len(buf) < 9:
+                buf += conn.recv(65536)
+            ln = int.from_bytes(buf[:3], "big")
+            ft, fl = buf[3], buf[4]
+            st = int.from_bytes(buf[5:9], "big") & 0x7FFFFFFF
+            while len(buf) < 9 + ln:
+                buf += conn.recv(65536)
+            pl = buf[9:9 + ln]
+            buf = buf[9 + ln:]
+            return ft, fl, st, pl
+
+        pre = b""
+        while len(pre) < len(PREFACE):
+            pre += conn.recv(len(PREFACE) - len(pre))
+        conn.sendall(frame(SETTINGS, 0, 0))
+        sid = None
+        while True:
+            ft, fl, st, pl = read_frame()
+            if ft == SETTINGS and not fl & ACK:
+                conn.sendall(frame(SETTINGS, ACK, 0))
+            elif ft == HEADERS:
+                sid = st
+            elif ft == DATA and st == sid and fl & END_STREAM:
+                break
+        if fail:
+            # trailers-only error: single HEADERS, END_STREAM, no body
+            conn.sendall(frame(
+                HEADERS, END_HEADERS | END_STREAM, sid,
+                enc.encode([(":status", "200"),
+                            ("content-type", "application/grpc"),
+                            ("grpc-status", "14"),
+                            ("grpc-message", "registry not ready")])))
+        else:
+            conn.sendall(frame(HEADERS, END_HEADERS, sid,
+                               enc.encode([(":status", "200"),
+                                           ("content-type",
+                                            "application/grpc")])))
+            conn.sendall(frame(DATA, 0, sid, grpc_frame(b"")))
+            conn.sendall(frame(HEADERS, END_HEADERS | END_STREAM, sid,
+                               enc.encode([("grpc-status", "0"),
+                                           ("grpc-message", "")])))
+            self.register_ok.set()
+
+    def _serve(self):
+        while not self.register_ok.is_set():
+            try:
+                conn, _ = self.srv.accept()
+            except OSError:
+                return
+            self.attempts += 1
+            try:
+                self._one(conn, fail=self.attempts == 1)
+            except Exception:
+                pass
+            finally:
+                conn.close()
+
+    def stop(self):
+        self.srv.close()
+
+
+def test_trailers_only_error_then_retry_succeeds(tmp_path):
+    kubelet_sock = str(tmp_path / "kubelet.sock")
+    fake = FlakyGrpcGoKubelet(kubelet_sock)
+    root = build_tree(tmp_path / "sys", n_gpus=1)
+    cfg = tmp_path / "config.yaml"
+    cfg.write_text(DEFAULT_CFG.format(replicas=1))
+    sock = str(tmp_path / "amd.sock")
+    proc = subprocess.Popen(
+        [str(PLUGIN), "--config", str(cfg), "--plugin-sock", sock,
+         "--kubelet-sock", kubelet_sock, "--register-retries", "5",
+         "--register-backoff-ms", "200", "--health-poll-ms", "0"],
+        env={"K3SAMD_SYSFS_ROOT": str(root)},
+        stdout=subprocess.PIPE, stderr=subprocess.PIPE)
+    try:
+        assert fake.register_ok.wait(15), "second Register never succeeded"
+        assert fake.attempts == 2
+        time.sleep(0.2)
+        assert proc.poll() is None  # serving after the retry
+    finally:
+        fake.stop()
+        proc.terminate()
+        proc.wait(timeout=10)
+        err = proc.stderr.read().decode()
+    assert "grpc=14" in err  # the trailers-only error surfaced in the log
