@@ -32,6 +32,7 @@
 
 #include "../common/json_writer.h"
 #include "../common/version.h"
+#include "../topology/gpu_health.h"
 #include "../topology/kfd_topology.h"
 
 namespace {
@@ -70,6 +71,14 @@ std::map<std::string, std::string> compute_labels(
   labels["amd.com/gpu.cu-count"] = std::to_string(g0.compute_units());
   if (!topo.driver_version.empty())
     labels["amd.com/gpu.driver-version"] = sanitize(topo.driver_version);
+  // RAS capability: ECC/error monitoring available (drives the health
+  // model's sick-GPU draining) — lets ECC-sensitive workloads select
+  // monitored nodes
+  {
+    k3samd::GpuHealthCounters hc = k3samd::read_gpu_health(
+        k3samd::default_sysfs_root(), g0.card_index);
+    labels["amd.com/gpu.ras"] = hc.ras_present ? "true" : "false";
+  }
   return labels;
 }
 

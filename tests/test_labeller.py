@@ -90,3 +90,13 @@ def test_cpu_only_no_labels(tmp_path):
     assert ff.read_text() == ""  # empty feature file => no gpu labels
     out = run_labeller(root, "--json")
     assert json.loads(out.stdout) == {}
+
+
+def test_ras_capability_label(tmp_path):
+    from sysfs_builder import make_gpu_sick
+    root = build_tree(tmp_path / "sys", n_gpus=1)
+    labels = json.loads(run_labeller(root, "--json").stdout)
+    assert labels["amd.com/gpu.ras"] == "false"  # fixture has no RAS dir
+    make_gpu_sick(root, 0, aca=True)  # creates ras/aca_* (all zeros)
+    labels = json.loads(run_labeller(root, "--json").stdout)
+    assert labels["amd.com/gpu.ras"] == "true"
