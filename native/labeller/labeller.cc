@@ -18,6 +18,7 @@
 //   amd.com/gpu.product=AMD-Instinct-MI355X
 //   amd.com/gpu.vram=294912Mi       amd.com/gpu.xgmi-links=7
 //   amd.com/gpu.cu-count=256        amd.com/gpu.driver-version=<ver>
+//   amd.com/gpu.ras=true|false      (ECC/RAS monitoring available)
 //
 // Modes: --oneshot [--json] (print and exit), daemon (rewrite every
 // --interval-s, default 60, picking up hotplug/health changes).
