@@ -113,3 +113,70 @@ def test_survives_many_half_open(plugin):
     for s in socks:
         s.close()
     assert_alive(sock, proc)
+
+
+def test_structured_frame_fuzz(plugin):
+    """Seeded random VALID-ish frame interleavings (settings, pings,
+    window updates, priority, padded headers/data fragments, rst,
+    unknown types) — the server must never crash and must still serve a
+    clean connection afterwards."""
+    import random
+    import struct as struct_mod
+    from grpcgo_wire import (GoHpackEncoder, PREFACE, frame, grpc_frame,
+                             SETTINGS, PING, WINDOW_UPDATE, HEADERS, DATA,
+                             RST_STREAM, END_HEADERS, END_STREAM)
+    sock_path, _proc = plugin
+    rng = random.Random(1234)
+    for trial in range(8):
+        s = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+        s.connect(sock_path)
+        s.settimeout(5)
+        out = bytearray(PREFACE + frame(SETTINGS, 0, 0))
+        enc = GoHpackEncoder()
+        sid = 1
+        for _ in range(rng.randint(3, 25)):
+            k = rng.randint(0, 7)
+            if k == 0:
+                out += frame(SETTINGS, 0, 0,
+                             struct_mod.pack(">HI", rng.randint(1, 6),
+                                             rng.randint(0, 1 << 20)))
+            elif k == 1:
+                out += frame(PING, 0, 0, bytes(rng.randrange(256)
+                                               for _ in range(8)))
+            elif k == 2:
+                out += frame(WINDOW_UPDATE, 0, rng.choice([0, sid]),
+                             struct_mod.pack(">I", rng.randint(1, 1 << 20)))
+            elif k == 3:
+                block = enc.encode([(":method", "POST"), (":scheme", "http"),
+                                    (":path",
+                                     "/v1beta1.DevicePlugin/GetDevicePluginOptions"),
+                                    (":authority", "x"),
+                                    ("content-type", "application/grpc"),
+                                    ("te", "trailers")])
+                out += frame(HEADERS, END_HEADERS, sid, block)
+                out += frame(DATA, END_STREAM, sid, grpc_frame(b""))
+                sid += 2
+            elif k == 4:
+                out += frame(RST_STREAM, 0, max(1, sid - 2),
+                             struct_mod.pack(">I", rng.randint(0, 13)))
+            elif k == 5:
+                out += frame(2, 0, sid, bytes(5))  # PRIORITY
+            elif k == 6:
+                out += frame(rng.randint(10, 200), rng.randrange(256),
+                             rng.choice([0, sid]),
+                             bytes(rng.randrange(64)))  # unknown type
+            else:
+                out += frame(DATA, 0, 999, b"orphan")  # unknown stream
+        try:
+            s.sendall(bytes(out))
+            s.recv(4096)
+        except OSError:
+            pass
+        s.close()
+    # the plugin survived all trials and serves a clean connection
+    from grpcgo_wire import GrpcGoConn
+    import pb_v1beta1 as pb
+    c = GrpcGoConn(sock_path)
+    assert pb.decode_options(
+        c.unary("/v1beta1.DevicePlugin/GetDevicePluginOptions"))
+    c.close()
