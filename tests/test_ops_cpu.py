@@ -40,3 +40,16 @@ def test_native_rejects_cpu_tensors():
     a = torch.zeros(16)
     with pytest.raises(Exception):
         ops.stream_triad(a, a.clone(), a.clone(), 1.0)
+
+
+def test_mx_gemm16_argument_validation():
+    """Binding-level checks fire before any launch (CPU-testable)."""
+    import pytest
+    import torch
+    from k3samd import ops
+    if not ops.native_available():
+        pytest.skip("extension not built")
+    a = torch.zeros(16, 128, dtype=torch.uint8)
+    b = torch.zeros(16, 128, dtype=torch.uint8)
+    with pytest.raises(RuntimeError, match="GPU"):
+        ops.mx_gemm16(a, b, 0)  # CPU tensors rejected
