@@ -102,3 +102,45 @@ def test_mi_burn_job_contract():
     c = pod["containers"][0]
     assert c["resources"]["limits"]["amd.com/gpu"] == "1"
     assert "--burn" in c["command"]
+
+
+def test_generated_config_property(tmp_path):
+    """Property: random valid config documents (PyYAML-dumped, so any
+    quoting/indent style yaml.dump picks) parse in the C++ plugin with
+    the same semantics PyYAML sees — replicas drive fan-out, rename
+    drives the advertised name."""
+    import random
+    from sysfs_builder import build_tree
+    rng = random.Random(99)
+    root = build_tree(tmp_path / "sys", n_gpus=2)
+    for trial in range(12):
+        replicas = rng.choice([1, 2, 3, 4, 7, 16, 64, 256])
+        rename = rng.choice([True, False])
+        doc = {
+            "version": "v1",
+            "flags": {"migStrategy": "none"},
+            "sharing": {"timeSlicing": {
+                "renameByDefault": rename,
+                "failRequestsGreaterThanOne": rng.choice([True, False]),
+                "resources": [{"name": "amd.com/gpu",
+                               "replicas": replicas}],
+            }},
+            "health": {
+                "maxUncorrectableErrors": rng.choice([0, 5, -1]),
+                "maxCorrectableErrors": rng.choice([100, 10000, -1]),
+                "maxResets": rng.choice([0, -1]),
+            },
+        }
+        cfg = tmp_path / f"cfg{trial}.yaml"
+        cfg.write_text(yaml.dump(doc,
+                                 default_flow_style=rng.choice([False]) ))
+        out = subprocess.run(
+            [str(PLUGIN), "--config", str(cfg), "--oneshot"],
+            env={"K3SAMD_SYSFS_ROOT": str(root)},
+            capture_output=True, text=True, timeout=60)
+        assert out.returncode == 0, (doc, out.stderr)
+        j = json.loads(out.stdout)
+        assert j["allocatable"] == 2 * replicas, (doc, j)
+        expect_name = "amd.com/gpu.shared" if (rename and replicas > 1) \
+            else "amd.com/gpu"
+        assert j["resource"] == expect_name, (doc, j)
