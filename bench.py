@@ -227,11 +227,20 @@ def main(argv=None) -> int:
 
     t = torch.tensor([elapsed], dtype=torch.float64, device=device)
     if dist is not None:
+        # keep every rank's own time too: a straggler shows up as one
+        # low per-rank rate, not just a depressed aggregate
+        all_t = [torch.zeros_like(t) for _ in range(world_size)]
+        dist.all_gather(all_t, t)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
     t_max = float(t[0])
 
     value = n_gpus * step_bytes * args.steps / t_max / 1e9
     ms_per_step = t_max / args.steps * 1e3
+    if dist is not None and rank == 0:
+        per_rank = [round(step_bytes * args.steps / float(x[0]) / 1e9, 1)
+                    for x in all_t]
+        print(f"# per-rank GB/s: {per_rank} (straggler = lowest)",
+              file=sys.stderr)
 
     # ---- RCCL transport self-diagnosis (N>1 on the GPU path) ----
     transports = None
