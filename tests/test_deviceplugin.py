@@ -572,3 +572,28 @@ def test_register_gives_up_after_retries(tmp_path):
         capture_output=True, text=True, timeout=60)
     assert proc.returncode == 1
     assert "Register retry 2/2" in proc.stderr
+
+
+def test_health_threshold_hot_reload(harness, tmp_path):
+    """Tightening the health: thresholds via ConfigMap hot-reload drains
+    an already-erroring GPU without a plugin restart."""
+    from sysfs_builder import make_gpu_sick
+    lenient = DEFAULT_CFG.format(replicas=1) + (
+        "health:\n  maxCorrectableErrors: 1000\n  maxResets: -1\n")
+    h = harness(n_gpus=1, replicas=1, register=False, health_poll_ms=100,
+                cfg_text=lenient)
+    stream = h.stream("ListAndWatch", timeout=30)
+    assert pb.decode_list_and_watch(next(stream))[0]["health"] == "Healthy"
+    make_gpu_sick(h.root, card_index=0, umc_ce=500)  # under 1000: healthy
+    time.sleep(0.4)
+    cfg = tmp_path / "config.yaml"
+    time.sleep(0.05)
+    cfg.write_text(DEFAULT_CFG.format(replicas=1) + (
+        "health:\n  maxCorrectableErrors: 100\n  maxResets: -1\n"))
+    # reload pushes a device-list update; the sick GPU must now be drained
+    deadline = time.time() + 10
+    while time.time() < deadline:
+        devs = pb.decode_list_and_watch(next(stream))
+        if devs[0]["health"] == "Unhealthy":
+            break
+    assert devs[0]["health"] == "Unhealthy"
