@@ -621,3 +621,52 @@ def test_trailers_only_error_then_retry_succeeds(tmp_path):
         proc.wait(timeout=10)
         err = proc.stderr.read().decode()
     assert "grpc=14" in err  # the trailers-only error surfaced in the log
+
+
+def test_conformance_dance_under_asan(tmp_path):
+    """The same grpc-go dance against the AddressSanitizer build —
+    memory safety of the whole serving path (HPACK huffman/dynamic
+    table, frame reassembly, flow control) under the strictest client."""
+    asan = PLUGIN.parent / "k3samd-device-plugin-asan"
+    if not asan.exists():
+        subprocess.run(["make", "-C", str(REPO / "native"), "asan-plugin"],
+                       check=True, capture_output=True, timeout=600)
+    root = build_tree(tmp_path / "sys", n_gpus=2)
+    cfg = tmp_path / "config.yaml"
+    cfg.write_text(DEFAULT_CFG.format(replicas=2))
+    sock = str(tmp_path / "amd.sock")
+    proc = subprocess.Popen(
+        [str(asan), "--config", str(cfg), "--plugin-sock", sock,
+         "--no-register", "--health-poll-ms", "50"],
+        env={"K3SAMD_SYSFS_ROOT": str(root),
+             "ASAN_OPTIONS": "exitcode=66 detect_leaks=0"},
+        stdout=subprocess.PIPE, stderr=subprocess.PIPE)
+    try:
+        deadline = time.time() + 15
+        while not Path(sock).exists():
+            assert time.time() < deadline and proc.poll() is None
+            time.sleep(0.05)
+        c = GrpcGoConn(sock)
+        lw = c.start_stream("/v1beta1.DevicePlugin/ListAndWatch")
+        assert c.read_stream_event(lw)[0] == "headers"
+        kind, data = c.read_stream_event(lw, bdp_ping_on_data=True)
+        assert kind == "data"
+        assert c.wait_ping_ack(BDP_PING)
+        devs = pb.decode_list_and_watch(data[5:])
+        resp = pb.decode_allocate_response(
+            c.unary("/v1beta1.DevicePlugin/Allocate",
+                    pb.encode_allocate_request([[devs[0]["id"]]])))
+        assert resp[0]["envs"]["K3SAMD_RENDER_MINORS"]
+        c.cancel(lw)
+        c.goaway_and_close()
+        time.sleep(0.3)
+    finally:
+        proc.terminate()
+        try:
+            proc.wait(timeout=15)
+        except subprocess.TimeoutExpired:
+            proc.kill()
+            proc.wait()
+    err = proc.stderr.read().decode(errors="replace")
+    assert "AddressSanitizer" not in err, err[-3000:]
+    assert proc.returncode != 66, "ASan reported memory errors"
