@@ -478,9 +478,11 @@ def test_conformance_dance_under_tsan(tmp_path):
     cfg = tmp_path / "config.yaml"
     cfg.write_text(DEFAULT_CFG.format(replicas=2))
     sock = str(tmp_path / "amd.sock")
+    metrics_sock = str(tmp_path / "metrics.sock")
     proc = subprocess.Popen(
         [str(tsan), "--config", str(cfg), "--plugin-sock", sock,
-         "--no-register", "--health-poll-ms", "50"],
+         "--no-register", "--health-poll-ms", "50",
+         "--metrics-addr", f"unix:{metrics_sock}"],
         env={"K3SAMD_SYSFS_ROOT": str(root),
              "TSAN_OPTIONS": "exitcode=66 halt_on_error=0"},
         stdout=subprocess.PIPE, stderr=subprocess.PIPE)
@@ -496,9 +498,20 @@ def test_conformance_dance_under_tsan(tmp_path):
         assert c.wait_ping_ack(BDP_PING)
         for _ in range(3):
             c.unary("/v1beta1.DevicePlugin/GetDevicePluginOptions")
+            # concurrent metrics scrapes + a silent client while RPCs run
+            # (per-connection metric threads under the race detector)
+            ms = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+            ms.connect(metrics_sock)
+            ms.sendall(b"GET /metrics HTTP/1.0\r\n\r\n")
+            while ms.recv(65536):
+                pass
+            ms.close()
+        silent = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+        silent.connect(metrics_sock)  # never sends; 2 s timeout path
         c.cancel(lw)
         c.goaway_and_close()
         time.sleep(0.3)
+        silent.close()
     finally:
         proc.terminate()
         try:
