@@ -205,6 +205,7 @@ int main(int argc, char** argv) {
   }
 
   if (all_gpus) {
+    if (do_check && !self_check()) return 3;  // oracle on device 0 first
     // concurrent non-temporal triad on every visible GPU (the in-pod
     // demonstration of the headline metric at amd.com/gpu: N): per-GPU
     // buffers + stream, launches overlapped, one global wall-clock.
